@@ -1,0 +1,216 @@
+#!/usr/bin/env python3
+# -*- coding: utf-8 -*-
+"""Flagship benchmark: ResNet-50 DDP bf16 samples/sec (BASELINE.json metric).
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+launched for N>1 as one rank per GPU via torch.distributed.run (reads
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from env).  Does W untimed warmup steps,
+then times EXACTLY K steps bracketed by barrier + torch.cuda.synchronize on
+both sides, takes the MAX elapsed over ranks, and rank 0 prints ONE JSON line.
+
+Synthetic data (ImageNet shape 3x224x224, random labels), random-init
+weights, bf16 autocast compute, full optimizer step in the timed region.
+
+Other configs (our own scaling table, not the driver default):
+    --model gpt2-oss    GPT-2-medium, OSS (ZeRO-1) shard
+    --model llama-fsdp  Llama-3-8B shape, FSDP (ZeRO-3) bf16
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=0, help="per-GPU batch (0=auto)")
+    p.add_argument("--model", type=str, default="resnet50",
+                   choices=["resnet50", "gpt2-oss", "llama-fsdp"])
+    p.add_argument("--seq", type=int, default=0, help="seq len for LM benches")
+    p.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
+    p.add_argument("--bucket-mb", type=int, default=64)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    distributed = world_size > 1
+
+    from stoke import Stoke, StokeOptimizer, DDPConfig
+    from stoke.ops.fused_adam import FusedAdamW
+    from benchmarks import models
+
+    torch.manual_seed(1234 + rank)
+    if args.cpu:
+        model = models.resnet18(num_classes=10, small_input=True)
+        batch = args.batch or 16
+        stoke_kw = dict(gpu=False, fp16=None, distributed=None)
+        data_shape = (batch, 3, 32, 32)
+        nclass, seq = 10, None
+        dtype = "fp32"
+        mname = "resnet18-cifar10-shape-cpu"
+    elif args.model == "resnet50":
+        torch.backends.cudnn.benchmark = True
+        model = models.resnet50(num_classes=1000)
+        batch = args.batch or 256
+        stoke_kw = dict(
+            gpu=True,
+            fp16="bf16",
+            distributed="ddp" if distributed else None,
+            configs=[DDPConfig(local_rank=int(os.environ.get("LOCAL_RANK", 0)),
+                               bucket_cap_mb=args.bucket_mb)],
+        )
+        data_shape = (batch, 3, 224, 224)
+        nclass, seq = 1000, None
+        dtype = "bf16"
+        mname = "resnet50"
+    elif args.model == "gpt2-oss":
+        seq = args.seq or 1024
+        model = models.gpt2_medium(max_seq=seq)
+        batch = args.batch or 16
+        stoke_kw = dict(
+            gpu=True,
+            fp16="bf16",
+            distributed="ddp" if distributed else None,
+            fairscale_oss=distributed,
+            configs=[DDPConfig(local_rank=int(os.environ.get("LOCAL_RANK", 0)),
+                               bucket_cap_mb=args.bucket_mb)],
+        )
+        data_shape = (batch, seq)
+        nclass = 50257
+        dtype = "bf16"
+        mname = "gpt2-medium"
+    else:  # llama-fsdp
+        seq = args.seq or 4096
+        model = models.llama3_8b(max_seq=seq)
+        batch = args.batch or 4
+        stoke_kw = dict(
+            gpu=True,
+            fp16="bf16",
+            distributed="ddp" if distributed else None,
+            fairscale_fsdp=distributed,
+            configs=[DDPConfig(local_rank=int(os.environ.get("LOCAL_RANK", 0)))],
+        )
+        data_shape = (batch, seq)
+        nclass = 128256
+        dtype = "bf16"
+        mname = "llama3-8b"
+
+    if args.model in ("gpt2-oss", "llama-fsdp"):
+        def loss_fn(logits, target):
+            return torch.nn.functional.cross_entropy(
+                logits.reshape(-1, logits.shape[-1]).float(), target.reshape(-1)
+            )
+    else:
+        loss_fn = torch.nn.CrossEntropyLoss()
+
+    s = Stoke(
+        model=model,
+        optimizer=StokeOptimizer(optimizer=FusedAdamW,
+                                 optimizer_kwargs={"lr": 1e-3}),
+        loss=loss_fn,
+        batch_size_per_device=batch,
+        verbose=False,
+        **stoke_kw,
+    )
+    device = torch.device("cpu") if args.cpu else torch.device(
+        "cuda", int(os.environ.get("LOCAL_RANK", 0)))
+
+    # Synthetic data: a few pre-generated batches cycled through
+    nbuf = 4
+    if seq is None:
+        xs = [torch.randn(*data_shape, device=device) for _ in range(nbuf)]
+        if not args.cpu:
+            xs = [x.to(memory_format=torch.channels_last) for x in xs]
+            s.model_access.to(memory_format=torch.channels_last)
+        ys = [torch.randint(0, nclass, (data_shape[0],), device=device)
+              for _ in range(nbuf)]
+    else:
+        xs = [torch.randint(0, nclass, data_shape, device=device)
+              for _ in range(nbuf)]
+        ys = [torch.randint(0, nclass, data_shape, device=device)
+              for _ in range(nbuf)]
+
+    def one_step(i):
+        x, y = xs[i % nbuf], ys[i % nbuf]
+        out = s.model(x)
+        loss = s.loss(out, y)
+        s.backward(loss)
+        s.step()
+
+    for i in range(args.warmup):
+        one_step(i)
+
+    if not args.cpu:
+        torch.cuda.synchronize()
+    s.barrier()
+    if not args.cpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    if not args.cpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    s.barrier()
+    if not args.cpu:
+        torch.cuda.synchronize()
+
+    # MAX elapsed over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device if not args.cpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world_size if distributed else (0 if args.cpu else 1)
+    samples = batch * max(n_gpus, 1) * args.steps
+    if seq is None:
+        value = samples / elapsed
+        unit = "samples/sec"
+        metric = f"samples/sec {mname} {'DDP ' if distributed else ''}{dtype}"
+        cfg_extra = {}
+    else:
+        value = samples * seq / elapsed
+        unit = "tokens/sec"
+        metric = f"tokens/sec {mname} {dtype}"
+        cfg_extra = {"seq_len": seq}
+
+    if rank == 0:
+        par = "dp%d" % world_size if distributed else ("cpu" if args.cpu else "single")
+        if args.model == "gpt2-oss" and distributed:
+            par = "dp%d+oss" % world_size
+        if args.model == "llama-fsdp" and distributed:
+            par = "fsdp%d" % world_size
+        print(json.dumps({
+            "metric": metric,
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": mname,
+                "global_batch": batch * max(n_gpus, 1),
+                "seq_len": cfg_extra.get("seq_len", 224),
+                "parallelism": par,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

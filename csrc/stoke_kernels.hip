@@ -1,0 +1,419 @@
+// stoke-amd native HIP kernels for MI355X (gfx950, CDNA4).
+//
+// These replace the CUDA kernels the reference implicitly relied on through
+// its dependencies:
+//   * torch.cuda.amp GradScaler ops (_amp_foreach_non_finite_check_and_unscale_,
+//     _amp_update_scale_)            -> multi_tensor_unscale / amp_update_scale
+//   * apex/fairscale FusedAdam      -> multi_tensor_adamw (+ bf16-param variant)
+//   * torch clip_grad_norm_ / _value_ reductions -> multi_tensor_l2norm / clamp
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//   * wavefront = 64; BLOCK = 256 threads (multiple of 64).
+//   * All ops here are HBM-bandwidth-bound elementwise/reduction work: the
+//     lever is vectorized 16 B/lane access (float4), not MFMA.
+//   * Multi-tensor-apply: one launch covers many tensors via a by-value
+//     kernarg struct of chunk descriptors (no per-tensor launches, no device
+//     metadata allocations).
+//   * Reductions do wave-level __shfl_down then one LDS step, one atomic per
+//     block (Guideline 12).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <vector>
+
+#define BLOCK 256
+#define CHUNK (1 << 16)  // elements per block-chunk
+#define MAX_TENSORS 24
+#define MAX_BLOCKS 256
+
+namespace {
+
+// ---------------------------------------------------------------------------
+// Multi-tensor-apply plumbing: chunk descriptors passed by value in kernarg.
+// ---------------------------------------------------------------------------
+template <int DEPTH>
+struct ChunkMeta {
+  const void* addr[DEPTH][MAX_TENSORS];
+  int sizes[MAX_TENSORS];
+  unsigned char block_to_tensor[MAX_BLOCKS];
+  int block_to_chunk[MAX_BLOCKS];
+  int nblocks;
+};
+
+template <int DEPTH, typename Functor, typename... Args>
+__global__ __launch_bounds__(BLOCK) void mta_kernel(ChunkMeta<DEPTH> meta,
+                                                    Functor fn, Args... args) {
+  const int bid = blockIdx.x;
+  if (bid >= meta.nblocks) return;
+  const int t = meta.block_to_tensor[bid];
+  const int chunk = meta.block_to_chunk[bid];
+  const int n = meta.sizes[t];
+  const int start = chunk * CHUNK;
+  const int len = min(CHUNK, n - start);
+  fn(meta, t, start, len, args...);
+}
+
+template <int DEPTH, typename Functor, typename... Args>
+void multi_tensor_apply(const std::vector<std::vector<at::Tensor>>& lists,
+                        Functor fn, Args... args) {
+  TORCH_CHECK(lists.size() == DEPTH, "depth mismatch");
+  const int ntensors = lists[0].size();
+  if (ntensors == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  ChunkMeta<DEPTH> meta;
+  // Find the last non-empty tensor so the final launch triggers in-loop.
+  int last_idx = -1;
+  for (int i = 0; i < ntensors; ++i)
+    if (lists[0][i].numel() > 0) last_idx = i;
+  if (last_idx < 0) return;
+  int ti = -1;  // tensor slot within meta
+  int bi = 0;   // block slot within meta
+  for (int i = 0; i < ntensors; ++i) {
+    const int n = lists[0][i].numel();
+    if (n == 0) continue;
+    ++ti;
+    meta.sizes[ti] = n;
+    for (int d = 0; d < DEPTH; ++d) meta.addr[d][ti] = lists[d][i].data_ptr();
+    const int nchunks = (n + CHUNK - 1) / CHUNK;
+    for (int c = 0; c < nchunks; ++c) {
+      meta.block_to_tensor[bi] = ti;
+      meta.block_to_chunk[bi] = c;
+      ++bi;
+      const bool tensors_full = (ti == MAX_TENSORS - 1 && c == nchunks - 1);
+      const bool blocks_full = (bi == MAX_BLOCKS);
+      const bool is_last = (i == last_idx && c == nchunks - 1);
+      if (tensors_full || blocks_full || is_last) {
+        meta.nblocks = bi;
+        hipLaunchKernelGGL((mta_kernel<DEPTH, Functor, Args...>), dim3(bi),
+                           dim3(BLOCK), 0, stream.stream(), meta, fn, args...);
+        bi = 0;
+        if (c == nchunks - 1) {
+          ti = -1;  // next tensor starts a fresh meta
+        } else {
+          // Mid-tensor flush: keep this tensor in slot 0 for remaining chunks.
+          meta.sizes[0] = n;
+          for (int d = 0; d < DEPTH; ++d) meta.addr[d][0] = meta.addr[d][ti];
+          ti = 0;
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Helpers
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float warp_reduce_sum(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off);
+  return v;
+}
+
+__device__ __forceinline__ float block_reduce_sum(float v) {
+  __shared__ float smem[BLOCK / 64];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  v = warp_reduce_sum(v);
+  if (lane == 0) smem[wid] = v;
+  __syncthreads();
+  v = (threadIdx.x < BLOCK / 64) ? smem[threadIdx.x] : 0.f;
+  if (wid == 0) {
+    #pragma unroll
+    for (int off = BLOCK / 128; off > 0; off >>= 1) v += __shfl_down(v, off);
+  }
+  return v;
+}
+
+typedef float float4v __attribute__((ext_vector_type(4)));
+
+// ---------------------------------------------------------------------------
+// Functors
+// ---------------------------------------------------------------------------
+
+// grads *= inv_scale; found_inf = 1.0 if any non-finite (fused unscale+check).
+struct UnscaleFunctor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             const float* inv_scale, float* found_inf) const {
+    float* g = ((float*)meta.addr[0][t]) + start;
+    const float inv = *inv_scale;
+    bool bad = false;
+    // float4 path only for 16-B-aligned chunk bases (grad views may be offset)
+    const int n4 = ((reinterpret_cast<uintptr_t>(g) & 15) == 0) ? (len & ~3) : 0;
+    for (int i = threadIdx.x * 4; i < n4; i += BLOCK * 4) {
+      float4v v = *(const float4v*)(g + i);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float x = v[k] * inv;
+        bad |= !isfinite(x);
+        v[k] = x;
+      }
+      *(float4v*)(g + i) = v;
+    }
+    for (int i = n4 + threadIdx.x; i < len; i += BLOCK) {
+      float x = g[i] * inv;
+      bad |= !isfinite(x);
+      g[i] = x;
+    }
+    if (__any(bad) && (threadIdx.x & 63) == 0) *found_inf = 1.0f;
+  }
+};
+
+// out[0] += sum(x*x) over all chunks (fp32 or bf16 input)
+template <typename T>
+struct L2NormFunctor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             float* out) const {
+    const T* g = ((const T*)meta.addr[0][t]) + start;
+    float acc = 0.f;
+    for (int i = threadIdx.x; i < len; i += BLOCK) {
+      float x = (float)g[i];
+      acc += x * x;
+    }
+    acc = block_reduce_sum(acc);
+    if (threadIdx.x == 0) atomicAdd(out, acc);
+  }
+};
+
+// x *= scale (by pointer so it composes with device-side clip coefficients)
+struct ScalePtrFunctor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             const float* scale) const {
+    float* g = ((float*)meta.addr[0][t]) + start;
+    const float s = *scale;
+    const int n4 = ((reinterpret_cast<uintptr_t>(g) & 15) == 0) ? (len & ~3) : 0;
+    for (int i = threadIdx.x * 4; i < n4; i += BLOCK * 4) {
+      float4v v = *(const float4v*)(g + i);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) v[k] *= s;
+      *(float4v*)(g + i) = v;
+    }
+    for (int i = n4 + threadIdx.x; i < len; i += BLOCK) g[i] *= s;
+  }
+};
+
+// clamp to [-limit, limit]
+struct ClampFunctor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             float limit) const {
+    float* g = ((float*)meta.addr[0][t]) + start;
+    const int n4 = ((reinterpret_cast<uintptr_t>(g) & 15) == 0) ? (len & ~3) : 0;
+    for (int i = threadIdx.x * 4; i < n4; i += BLOCK * 4) {
+      float4v v = *(const float4v*)(g + i);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) v[k] = fminf(fmaxf(v[k], -limit), limit);
+      *(float4v*)(g + i) = v;
+    }
+    for (int i = n4 + threadIdx.x; i < len; i += BLOCK)
+      g[i] = fminf(fmaxf(g[i], -limit), limit);
+  }
+};
+
+// AdamW, fp32 params/grads/state, decoupled weight decay, bias correction.
+// Optionally skips the whole update when *found_inf != 0 (scaler-integrated),
+// and unscales the gradient in-register (grad_scale = 1/loss_scale).
+struct AdamWFunctor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             float lr, float beta1, float beta2, float eps,
+                             float weight_decay, float bc1, float bc2,
+                             const float* found_inf, const float* inv_scale) const {
+    if (found_inf && *found_inf != 0.f) return;
+    float* p = ((float*)meta.addr[0][t]) + start;
+    float* g = ((float*)meta.addr[1][t]) + start;
+    float* m = ((float*)meta.addr[2][t]) + start;
+    float* v = ((float*)meta.addr[3][t]) + start;
+    const float inv = inv_scale ? *inv_scale : 1.f;
+    const int n4 = ((reinterpret_cast<uintptr_t>(p) & 15) == 0 &&
+                    (reinterpret_cast<uintptr_t>(g) & 15) == 0 &&
+                    (reinterpret_cast<uintptr_t>(m) & 15) == 0 &&
+                    (reinterpret_cast<uintptr_t>(v) & 15) == 0)
+                       ? (len & ~3) : 0;
+    for (int i = threadIdx.x * 4; i < n4; i += BLOCK * 4) {
+      float4v pv = *(const float4v*)(p + i);
+      float4v gv = *(const float4v*)(g + i);
+      float4v mv = *(const float4v*)(m + i);
+      float4v vv = *(const float4v*)(v + i);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float gg = gv[k] * inv;
+        float mm = beta1 * mv[k] + (1.f - beta1) * gg;
+        float vn = beta2 * vv[k] + (1.f - beta2) * gg * gg;
+        float denom = sqrtf(vn / bc2) + eps;
+        float update = (mm / bc1) / denom + weight_decay * pv[k];
+        pv[k] = pv[k] - lr * update;
+        mv[k] = mm;
+        vv[k] = vn;
+      }
+      *(float4v*)(p + i) = pv;
+      *(float4v*)(m + i) = mv;
+      *(float4v*)(v + i) = vv;
+    }
+    for (int i = n4 + threadIdx.x; i < len; i += BLOCK) {
+      float gg = g[i] * inv;
+      float mm = beta1 * m[i] + (1.f - beta1) * gg;
+      float vn = beta2 * v[i] + (1.f - beta2) * gg * gg;
+      float denom = sqrtf(vn / bc2) + eps;
+      float update = (mm / bc1) / denom + weight_decay * p[i];
+      p[i] = p[i] - lr * update;
+      m[i] = mm;
+      v[i] = vn;
+    }
+  }
+};
+
+// AdamW with bf16 model params + fp32 master params/state, bf16 grads
+// (the FSDP bf16 flat-shard path: depth 5 = [bf16 p, bf16 g, fp32 m, fp32 v,
+// fp32 master]).
+struct AdamWBF16Functor {
+  template <typename Meta>
+  __device__ void operator()(const Meta& meta, int t, int start, int len,
+                             float lr, float beta1, float beta2, float eps,
+                             float weight_decay, float bc1, float bc2,
+                             const float* found_inf, const float* inv_scale) const {
+    if (found_inf && *found_inf != 0.f) return;
+    __hip_bfloat16* p = ((__hip_bfloat16*)meta.addr[0][t]) + start;
+    const __hip_bfloat16* g = ((const __hip_bfloat16*)meta.addr[1][t]) + start;
+    float* m = ((float*)meta.addr[2][t]) + start;
+    float* v = ((float*)meta.addr[3][t]) + start;
+    float* w = ((float*)meta.addr[4][t]) + start;  // fp32 master copy
+    const float inv = inv_scale ? *inv_scale : 1.f;
+    for (int i = threadIdx.x; i < len; i += BLOCK) {
+      float gg = __bfloat162float(g[i]) * inv;
+      float mm = beta1 * m[i] + (1.f - beta1) * gg;
+      float vn = beta2 * v[i] + (1.f - beta2) * gg * gg;
+      float denom = sqrtf(vn / bc2) + eps;
+      float pw = w[i];
+      float update = (mm / bc1) / denom + weight_decay * pw;
+      pw = pw - lr * update;
+      w[i] = pw;
+      p[i] = __float2bfloat16(pw);
+      m[i] = mm;
+      v[i] = vn;
+    }
+  }
+};
+
+// Dynamic loss-scale update with hysteresis (replaces _amp_update_scale_).
+__global__ void amp_update_scale_kernel(float* scale, int* growth_tracker,
+                                        const float* found_inf,
+                                        float growth_factor,
+                                        float backoff_factor,
+                                        int growth_interval) {
+  if (*found_inf != 0.f) {
+    *scale *= backoff_factor;
+    *growth_tracker = 0;
+  } else {
+    int g = *growth_tracker + 1;
+    if (g >= growth_interval) {
+      *scale *= growth_factor;
+      g = 0;
+    }
+    *growth_tracker = g;
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Python-visible entry points
+// ---------------------------------------------------------------------------
+
+void multi_tensor_unscale_(std::vector<at::Tensor> grads, at::Tensor inv_scale,
+                           at::Tensor found_inf) {
+  multi_tensor_apply<1>({grads}, UnscaleFunctor{},
+                        inv_scale.data_ptr<float>(),
+                        found_inf.data_ptr<float>());
+}
+
+at::Tensor multi_tensor_l2norm_sq(std::vector<at::Tensor> tensors) {
+  auto out = at::zeros({1}, tensors[0].options().dtype(at::kFloat));
+  if (tensors[0].scalar_type() == at::kBFloat16) {
+    multi_tensor_apply<1>({tensors}, L2NormFunctor<__hip_bfloat16>{},
+                          out.data_ptr<float>());
+  } else {
+    multi_tensor_apply<1>({tensors}, L2NormFunctor<float>{},
+                          out.data_ptr<float>());
+  }
+  return out;
+}
+
+void multi_tensor_scale_(std::vector<at::Tensor> tensors, at::Tensor scale) {
+  multi_tensor_apply<1>({tensors}, ScalePtrFunctor{}, scale.data_ptr<float>());
+}
+
+void multi_tensor_clamp_(std::vector<at::Tensor> tensors, double limit) {
+  multi_tensor_apply<1>({tensors}, ClampFunctor{}, (float)limit);
+}
+
+void multi_tensor_adamw_(std::vector<at::Tensor> params,
+                         std::vector<at::Tensor> grads,
+                         std::vector<at::Tensor> exp_avgs,
+                         std::vector<at::Tensor> exp_avg_sqs,
+                         int64_t step, double lr, double beta1, double beta2,
+                         double eps, double weight_decay,
+                         c10::optional<at::Tensor> found_inf,
+                         c10::optional<at::Tensor> inv_scale) {
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  const float* fi = found_inf ? found_inf->data_ptr<float>() : nullptr;
+  const float* is = inv_scale ? inv_scale->data_ptr<float>() : nullptr;
+  multi_tensor_apply<4>({params, grads, exp_avgs, exp_avg_sqs}, AdamWFunctor{},
+                        (float)lr, (float)beta1, (float)beta2, (float)eps,
+                        (float)weight_decay, bc1, bc2, fi, is);
+}
+
+void multi_tensor_adamw_bf16_(std::vector<at::Tensor> params,
+                              std::vector<at::Tensor> grads,
+                              std::vector<at::Tensor> exp_avgs,
+                              std::vector<at::Tensor> exp_avg_sqs,
+                              std::vector<at::Tensor> masters,
+                              int64_t step, double lr, double beta1,
+                              double beta2, double eps, double weight_decay,
+                              c10::optional<at::Tensor> found_inf,
+                              c10::optional<at::Tensor> inv_scale) {
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  const float* fi = found_inf ? found_inf->data_ptr<float>() : nullptr;
+  const float* is = inv_scale ? inv_scale->data_ptr<float>() : nullptr;
+  multi_tensor_apply<5>({params, grads, exp_avgs, exp_avg_sqs, masters},
+                        AdamWBF16Functor{}, (float)lr, (float)beta1,
+                        (float)beta2, (float)eps, (float)weight_decay, bc1,
+                        bc2, fi, is);
+}
+
+void amp_update_scale_(at::Tensor scale, at::Tensor growth_tracker,
+                       at::Tensor found_inf, double growth_factor,
+                       double backoff_factor, int64_t growth_interval) {
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(amp_update_scale_kernel, dim3(1), dim3(1), 0,
+                     stream.stream(), scale.data_ptr<float>(),
+                     growth_tracker.data_ptr<int>(),
+                     found_inf.data_ptr<float>(), (float)growth_factor,
+                     (float)backoff_factor, (int)growth_interval);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
+        "fused grad unscale + inf/nan check (HIP)");
+  m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,
+        "sum of squares over tensor list (HIP)");
+  m.def("multi_tensor_scale_", &multi_tensor_scale_,
+        "in-place scale by device scalar (HIP)");
+  m.def("multi_tensor_clamp_", &multi_tensor_clamp_,
+        "in-place clamp to [-limit, limit] (HIP)");
+  m.def("multi_tensor_adamw_", &multi_tensor_adamw_,
+        "fused AdamW, fp32 (HIP)");
+  m.def("multi_tensor_adamw_bf16_", &multi_tensor_adamw_bf16_,
+        "fused AdamW, bf16 params + fp32 master (HIP)");
+  m.def("amp_update_scale_", &amp_update_scale_,
+        "dynamic loss scale update (HIP)");
+  m.attr("_built_for") = "gfx950";
+}

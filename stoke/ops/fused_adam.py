@@ -1,0 +1,101 @@
+# -*- coding: utf-8 -*-
+"""Fused AdamW optimizer backed by the native HIP multi-tensor kernel.
+
+The MI355X-native replacement for the apex/deepspeed/fairscale FusedAdam the
+reference detects by name (``stoke/utils.py:103``):
+one kernel launch updates every parameter (per dtype group), with optional
+scaler integration (the kernel itself skips the step when found_inf is set,
+so no host sync is needed on the skip path).
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+from stoke import ops
+
+
+class FusedAdamW(torch.optim.Optimizer):
+    """AdamW with decoupled weight decay; multi-tensor fused on ROCm.
+
+    The class name contains "Fused" on purpose: zero_grad keeps gradients
+    allocated (reference ``utils.py:103-106`` semantics).
+    """
+
+    # StokeGradScaler passes found_inf to step() so the skip-on-overflow
+    # decision stays on device (no host sync).
+    step_supports_found_inf = True
+
+    def __init__(
+        self,
+        params,
+        lr: float = 1e-3,
+        betas=(0.9, 0.999),
+        eps: float = 1e-8,
+        weight_decay: float = 1e-2,
+        adam_w_mode: bool = True,
+    ):
+        if lr < 0.0:
+            raise ValueError(f"Invalid learning rate: {lr}")
+        defaults = dict(
+            lr=lr, betas=betas, eps=eps, weight_decay=weight_decay,
+            adam_w_mode=adam_w_mode,
+        )
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(
+        self,
+        closure=None,
+        found_inf: Optional[torch.Tensor] = None,
+        inv_scale: Optional[torch.Tensor] = None,
+    ):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            wd = group["weight_decay"] if group["adam_w_mode"] else 0.0
+            fp32_p, fp32_g, fp32_m, fp32_v = [], [], [], []
+            bf16_p, bf16_g, bf16_m, bf16_v, bf16_w = [], [], [], [], []
+            step_t = None
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                    if p.dtype == torch.bfloat16:
+                        state["master"] = p.detach().float()
+                state["step"] += 1
+                step_t = state["step"]
+                if p.dtype == torch.bfloat16:
+                    bf16_p.append(p.data)
+                    bf16_g.append(p.grad.data)
+                    bf16_m.append(state["exp_avg"])
+                    bf16_v.append(state["exp_avg_sq"])
+                    bf16_w.append(state["master"])
+                else:
+                    fp32_p.append(p.data)
+                    fp32_g.append(p.grad.data)
+                    fp32_m.append(state["exp_avg"])
+                    fp32_v.append(state["exp_avg_sq"])
+            if step_t is None:
+                continue
+            if fp32_p:
+                ops.fused_adamw_(
+                    fp32_p, fp32_g, fp32_m, fp32_v, step_t,
+                    group["lr"], beta1, beta2, group["eps"], wd,
+                    found_inf=found_inf, inv_scale=inv_scale,
+                )
+            if bf16_p:
+                ops.fused_adamw_(
+                    bf16_p, bf16_g, bf16_m, bf16_v, step_t,
+                    group["lr"], beta1, beta2, group["eps"], wd,
+                    found_inf=found_inf, inv_scale=inv_scale, masters=bf16_w,
+                )
+        return loss

@@ -1,0 +1,428 @@
+# -*- coding: utf-8 -*-
+"""The unified runtime runner.
+
+The reference builds its runtime by dynamically composing one mixin from each
+of four families (5 distributed x 5 fp16 x 2 optimizer x 4 io =
+``type("StokeRunner", ...)`` at ``stoke/stoke.py:599-657``).  With a single
+RCCL backend that matrix collapses: this ONE class implements the same runner
+protocol (``setup_distributed``, ``wrap_distributed``, ``wrap_fp16``,
+``build_optimizer``, ``backward_call``, ``step_call``, ``clip_grad``,
+``grad_accum_context``, ``step_context``, ``model_context``/``loss_context``,
+``detach_and_sync_loss``, ``print_device``, ``save``/``load``, ``barrier``,
+``scaler``, ``rank``, ``world_size``, ``device_id``, ``initialized``,
+``clean`` — SURVEY.md section 1.1) parameterized on three orthogonal axes:
+
+* device mode : cpu | gpu | distributed (one RCCL process group)
+* precision   : fp32 | bf16 autocast (scaler-free on CDNA4) | fp16 + native
+                dynamic scaler (covers "amp", "apex_O1/O2", "deepspeed")
+* shard level : none (DDP engine) | oss (ZeRO-1) | oss+sddp (ZeRO-2) |
+                fsdp (ZeRO-3); deepspeed ZeRO stages map onto the same levels
+"""
+
+from contextlib import nullcontext
+from typing import List, Optional, Tuple, Union
+
+import torch
+
+from stoke.amp import StokeGradScaler
+from stoke.comm import StokeProcessGroup
+from stoke.configs import ClipGradConfig, ClipGradNormConfig
+from stoke.ddp import StokeDDPModule
+from stoke.shard import OSSOptimizer, StokeFSDPModule, StokeSDDPModule
+from stoke import io_ops
+from stoke.utils import unrolled_print, zero_optimizer_grads
+
+
+class StokeRunner:
+    """Single runtime object backing the ``Stoke`` facade."""
+
+    def __init__(
+        self,
+        status,
+        loss,
+        verbose: bool = True,
+        info_rank: Union[int, List[int], None] = 0,
+    ):
+        self._status = status
+        self._loss = loss
+        self._verbose = verbose
+        self._info_rank = info_rank
+        self._pg: Optional[StokeProcessGroup] = None
+        self._scaler: Optional[StokeGradScaler] = None
+        self._engine = None  # DDP / SDDP / FSDP wrapper (has finish_backward)
+        self._device = torch.device("cpu")
+        self._grads_ready = False
+        self._ds_step_counter = 0
+        # Resolve the three axes from the status
+        s = status
+        if s.distributed is not None:
+            self._mode = "distributed"
+        elif s.gpu:
+            self._mode = "gpu"
+        else:
+            self._mode = "cpu"
+        if s.fully_sharded:
+            self._shard = "fsdp"
+        elif s.sharded:
+            self._shard = "sddp"
+        elif s.oss:
+            self._shard = "oss"
+        elif s.is_distributed_deepspeed and (s.zero or 0) > 0:
+            self._shard = {1: "oss", 2: "sddp", 3: "fsdp"}[min(s.zero, 3)]
+        else:
+            self._shard = "none"
+        fp16 = s.fp16
+        if fp16 is None:
+            self._precision = "fp32"
+        elif fp16 == "bf16":
+            self._precision = "bf16"
+        else:  # amp / apex_O1 / apex_O2 / deepspeed -> native fp16 + scaler
+            self._precision = "fp16"
+
+    # ------------------------------------------------------------- lifecycle
+    def setup_distributed(self):
+        s = self._status
+        if self._mode == "distributed":
+            if s.is_distributed_deepspeed:
+                cfg = s.deepspeed_config
+                backend, init_method = cfg.dist_backend, cfg.init_method
+                local_rank = None
+                auto_mpi = cfg.auto_mpi_discovery
+            else:
+                cfg = s.ddp_config
+                backend, init_method = cfg.backend, cfg.init_method
+                local_rank = cfg.local_rank
+                auto_mpi = cfg.auto_mpi_discovery
+            self._pg = StokeProcessGroup(
+                backend=backend,
+                init_method=init_method,
+                local_rank=local_rank,
+                auto_mpi_discovery=auto_mpi,
+            )
+            self._device = self._pg.device
+        elif self._mode == "gpu":
+            self._device = torch.device("cuda", torch.cuda.current_device())
+        else:
+            self._device = torch.device("cpu")
+        # Precision machinery
+        if self._precision == "fp16":
+            if self._status.is_fp16_deepspeed:
+                ds = self._status.deepspeed_config.fp16
+                init_scale = 2.0**ds.initial_scale_power if ds.loss_scale == 0.0 \
+                    else ds.loss_scale
+                self._scaler = StokeGradScaler(
+                    init_scale=init_scale,
+                    growth_factor=2.0,
+                    backoff_factor=0.5,
+                    growth_interval=ds.loss_scale_window,
+                    device=str(self._device),
+                    sharded=self._shard in ("sddp", "fsdp"),
+                )
+            else:
+                amp = self._status.amp_config
+                self._scaler = StokeGradScaler(
+                    init_scale=amp.init_scale,
+                    growth_factor=amp.growth_factor,
+                    backoff_factor=amp.backoff_factor,
+                    growth_interval=amp.growth_interval,
+                    device=str(self._device),
+                    sharded=self._shard in ("sddp", "fsdp"),
+                )
+
+    @property
+    def initialized(self) -> bool:
+        if self._mode == "distributed":
+            return self._pg is not None
+        return True
+
+    @property
+    def rank(self):
+        if self._mode == "cpu":
+            return "cpu"
+        if self._mode == "gpu":
+            return "gpu"
+        return self._pg.rank if self._pg else 0
+
+    @property
+    def world_size(self) -> int:
+        if self._mode == "distributed" and self._pg:
+            return self._pg.world_size
+        return 1
+
+    @property
+    def device_id(self):
+        if self._mode == "cpu":
+            return "cpu"
+        if self._mode == "gpu":
+            return torch.cuda.current_device()
+        return self._pg.local_rank if self._pg else 0
+
+    @property
+    def device(self) -> torch.device:
+        return self._device
+
+    def barrier(self):
+        if self._pg is not None:
+            self._pg.barrier()
+
+    def clean(self):
+        if self._pg is not None:
+            self._pg.clean()
+            self._pg = None
+
+    # ----------------------------------------------------------------- wraps
+    def wrap_distributed(
+        self, model: torch.nn.Module, grad_accum: int, optimizer=None
+    ) -> Tuple[torch.nn.Module, object]:
+        """Wrap the model with the right engine for the shard level."""
+        s = self._status
+        if self._mode != "distributed":
+            return model, optimizer
+        # SyncBN conversion happens before any engine wrapping
+        convert_bn = False
+        if s.is_distributed_ddp and s.ddp_config.convert_to_sync_batch_norm:
+            convert_bn = True
+        if s.is_distributed_horovod and s.horovod_config.convert_to_sync_batch_norm:
+            convert_bn = True
+        if convert_bn:
+            model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+        if self._shard == "fsdp":
+            fcfg = s.fsdp_config
+            wrapped = StokeFSDPModule(
+                model,
+                pg=self._pg,
+                compute_dtype=fcfg.compute_dtype,
+                mixed_precision=fcfg.mixed_precision or self._precision != "fp32",
+                fp32_reduce_scatter=fcfg.fp32_reduce_scatter,
+                reshard_after_forward=fcfg.reshard_after_forward,
+                disable_reshard_on_root=fcfg.disable_reshard_on_root,
+                flatten_parameters=fcfg.flatten_parameters,
+            )
+            self._engine = wrapped
+            return wrapped, optimizer
+        if self._shard == "sddp":
+            scfg = s.sddp_config
+            if optimizer is None:
+                raise RuntimeError(
+                    "Stoke -- SDDP wrap requires the OSS optimizer to exist first"
+                )
+            wrapped = StokeSDDPModule(
+                model,
+                sharded_optimizer=optimizer,
+                pg=self._pg,
+                broadcast_buffers=scfg.broadcast_buffers,
+                sync_models_at_startup=scfg.sync_models_at_startup,
+                reduce_buffer_size=scfg.reduce_buffer_size,
+                reduce_fp16=scfg.reduce_fp16,
+            )
+            self._engine = wrapped
+            return wrapped, optimizer
+        # Plain bucketed-all-reduce DDP engine (also the horovod/deepspeed
+        # stage-0/1 gradient path)
+        dcfg = s.ddp_config if not s.is_distributed_deepspeed else None
+        hv = s.horovod_config if s.is_distributed_horovod else None
+        wrapped = StokeDDPModule(
+            model,
+            pg=self._pg,
+            bucket_cap_mb=dcfg.bucket_cap_mb if dcfg else 64,
+            broadcast_buffers=dcfg.broadcast_buffers if dcfg else True,
+            gradient_as_bucket_view=dcfg.gradient_as_bucket_view if dcfg else False,
+            find_unused_parameters=dcfg.find_unused_parameters if dcfg else False,
+            compress_fp16=bool(hv.compression) if hv else False,
+            gradient_predivide_factor=hv.gradient_predivide_factor if hv else 1.0,
+            average_grads=(hv is None or hv.op == "Average"
+                           or getattr(hv.op, "value", hv.op) == "Average"),
+        )
+        self._engine = wrapped
+        return wrapped, optimizer
+
+    def wrap_fp16(self, model=None, optimizer=None):
+        if self._verbose and self._scaler is not None:
+            self.print_device(
+                f"FP16: native dynamic scaler (init_scale={self._scaler.get_scale()})",
+                rank=self._info_rank,
+            )
+        return model, optimizer
+
+    def build_optimizer(self, optimizer, optimizer_kwargs, model):
+        """Instantiate the optimizer (plain or OSS-sharded)."""
+        if self._shard in ("oss", "sddp"):
+            ocfg = self._status.oss_config
+            opt = OSSOptimizer(
+                [p for p in model.parameters() if p.requires_grad],
+                optim=optimizer,
+                pg=self._pg,
+                broadcast_fp16=ocfg.broadcast_fp16,
+                **optimizer_kwargs,
+            )
+        else:
+            params = [p for p in model.parameters() if p.requires_grad]
+            opt = optimizer(params, **optimizer_kwargs)
+        if self._verbose:
+            self.print_device(
+                f"Optimizer: {type(opt).__name__} "
+                f"(shard level: {self._shard})",
+                rank=self._info_rank,
+            )
+        return opt
+
+    # -------------------------------------------------------------- contexts
+    @property
+    def model_context(self):
+        if self._precision == "bf16":
+            return torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+        if self._precision == "fp16":
+            return torch.autocast(device_type="cuda", dtype=torch.float16)
+        return nullcontext()
+
+    @property
+    def loss_context(self):
+        return self.model_context
+
+    def grad_accum_context(self, model):
+        no_sync_ok = True
+        s = self._status
+        if s.is_distributed_ddp and not s.ddp_config.no_sync:
+            no_sync_ok = False
+        if no_sync_ok and hasattr(model, "no_sync"):
+            return model.no_sync()
+        return nullcontext()
+
+    def step_context(self, optimizer):
+        return nullcontext()
+
+    # ------------------------------------------------------- backward / step
+    def backward_call(self, loss, model, optimizer):
+        self._grads_ready = False
+        if self._scaler is not None:
+            scaled = self._scaler.scale(loss)
+        else:
+            scaled = loss
+        if isinstance(scaled, (list, tuple)):
+            for idx, val in enumerate(scaled):
+                val.backward(retain_graph=(idx == 0))
+        else:
+            scaled.backward()
+
+    def _ensure_grads_ready(self, model):
+        """Flush any pending bucket reductions (idempotent per step)."""
+        if self._grads_ready:
+            return
+        target = model
+        if self._engine is not None:
+            target = self._engine
+        if hasattr(target, "finish_backward"):
+            target.finish_backward()
+        self._grads_ready = True
+
+    def step_call(self, model, optimizer):
+        self._ensure_grads_ready(model)
+        if self._status.is_distributed_deepspeed:
+            # Deepspeed contract: the facade calls step every micro-batch and
+            # the engine owns accumulation (reference fp16.py:354-373).
+            self._ds_step_counter += 1
+            if self._ds_step_counter % max(self._status.grad_accum, 1) != 0:
+                return
+        if self._scaler is not None:
+            self._scaler.step(optimizer)
+            self._scaler.update()
+        else:
+            optimizer.step()
+        if self._status.is_distributed_deepspeed:
+            zero_optimizer_grads(optimizer)
+        self._grads_ready = False
+
+    # ------------------------------------------------------------------ clip
+    def clip_grad(
+        self,
+        grad_clip: Union[ClipGradConfig, ClipGradNormConfig],
+        model,
+        optimizer,
+        oss: bool = False,
+        horovod: bool = False,
+        deepspeed: bool = False,
+        fsdp: bool = False,
+    ):
+        from stoke import ops
+
+        self._ensure_grads_ready(model)
+        if self._scaler is not None:
+            self._scaler.unscale_(optimizer)
+        # DeepSpeed-mode ZeRO levels behave like their fairscale-style twins
+        oss = oss or self._shard in ("oss", "sddp")
+        fsdp = fsdp or self._shard == "fsdp"
+        if isinstance(grad_clip, ClipGradConfig):
+            grads = [
+                p.grad
+                for group in optimizer.param_groups
+                for p in group["params"]
+                if p.grad is not None
+            ]
+            ops.multi_tensor_clamp_(grads, grad_clip.clip_value)
+        elif isinstance(grad_clip, ClipGradNormConfig):
+            if fsdp:
+                model.clip_grad_norm_(
+                    max_norm=grad_clip.max_norm, norm_type=grad_clip.norm_type
+                )
+            elif oss and isinstance(optimizer, OSSOptimizer):
+                optimizer.clip_grad_norm(
+                    max_norm=grad_clip.max_norm,
+                    norm_type=grad_clip.norm_type,
+                    grads_sharded=(self._shard == "sddp"),
+                )
+            else:
+                grads = [
+                    p.grad
+                    for group in optimizer.param_groups
+                    for p in group["params"]
+                    if p.grad is not None
+                ]
+                if grad_clip.norm_type == 2.0 and (
+                    not grads or grads[0].dtype == torch.float32
+                ):
+                    total = ops.multi_tensor_l2norm(grads)
+                    coef = torch.clamp(
+                        grad_clip.max_norm / (total + 1e-6), max=1.0
+                    )
+                    ops.multi_tensor_scale_(grads, coef.float())
+                else:
+                    torch.nn.utils.clip_grad_norm_(
+                        (p for g in optimizer.param_groups for p in g["params"]),
+                        max_norm=grad_clip.max_norm,
+                        norm_type=grad_clip.norm_type,
+                    )
+        else:
+            raise ValueError(
+                f"Stoke -- clip_grad received an incorrect instance type of {type(grad_clip)}"
+            )
+
+    # ------------------------------------------------------------- loss sync
+    def detach_and_sync_loss(self, loss, device=None):
+        if isinstance(loss, (list, tuple)):
+            return type(loss)(self.detach_and_sync_loss(v, device) for v in loss)
+        if self._mode == "distributed" and self._pg is not None:
+            return self._pg.sync_loss(loss, device)
+        return float(loss.detach().item())
+
+    # ---------------------------------------------------------------- print
+    def print_device(self, msg, rank=0, single_line: bool = False):
+        my_rank = self.rank
+        if isinstance(my_rank, str):
+            unrolled_print(msg, single_line=single_line)
+            return
+        ranks = rank if isinstance(rank, (list, tuple)) else [rank]
+        if my_rank in ranks or rank is None:
+            unrolled_print(msg, single_line=single_line)
+
+    # ------------------------------------------------------------------- IO
+    @property
+    def scaler(self):
+        return self._scaler
+
+    def save(self, **kwargs):
+        return io_ops.save_checkpoint(
+            runner=self, shard=self._shard, verbose=self._verbose, **kwargs
+        )
+
+    def load(self, **kwargs):
+        return io_ops.load_checkpoint(runner=self, shard=self._shard, **kwargs)
