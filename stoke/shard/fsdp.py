@@ -459,13 +459,17 @@ class StokeFSDPModule(nn.Module):
                 u = next((x for x in self.units if x.shard is shard_p), None)
                 if u is None:
                     continue
-                # Collect per-key flats over this unit's params
+                # Collect per-key flats over this unit's params.  Only state
+                # tensors shaped like the parameter (exp_avg, exp_avg_sq, ...)
+                # re-flatten; scalar tensors (e.g. Adam's `step`) pass through.
                 keys = set()
-                for name in u.param_names:
+                for p, name in zip(u.params, u.param_names):
                     st = full["state"].get(name_list.index(name))
                     if st:
                         keys.update(
-                            k for k, v in st.items() if isinstance(v, torch.Tensor)
+                            k for k, v in st.items()
+                            if isinstance(v, torch.Tensor)
+                            and v.numel() == p.numel() and p.numel() > 1
                         )
                 state_entry = {}
                 for k in keys:
@@ -486,6 +490,11 @@ class StokeFSDPModule(nn.Module):
                 # Scalar entries (e.g. step) come from the first param
                 st0 = full["state"].get(name_list.index(u.param_names[0]), {})
                 for k, v in st0.items():
-                    if not isinstance(v, torch.Tensor):
+                    if k in state_entry:
+                        continue
+                    if isinstance(v, torch.Tensor):
+                        if v.numel() == 1:
+                            state_entry[k] = v.clone()
+                    else:
                         state_entry[k] = v
                 optimizer.state[shard_p] = state_entry
