@@ -10,11 +10,13 @@ runtime (``stoke/runtime/runner.py``) instead of the reference's dynamic
 mixin composition.
 """
 
-from contextlib import nullcontext
+import os
+from contextlib import contextmanager, nullcontext
 from typing import Callable, Dict, List, Optional, Sequence, Tuple, Type, Union
 from uuid import uuid4
 
 import torch
+
 from torch.utils.data import Dataset
 from torch.utils.data.distributed import DistributedSampler, Sampler
 
@@ -43,6 +45,23 @@ from stoke.utils import (
     _worker_init_fn_t,
     zero_optimizer_grads,
 )
+
+
+# rocprofv3/roctracer-visible phase markers: enabled
+# with STOKE_NVTX=1; torch.cuda.nvtx maps to roctx ranges on ROCm.
+_NVTX = os.environ.get("STOKE_NVTX", "0") == "1"
+
+
+@contextmanager
+def _phase(name: str):
+    if _NVTX and torch.cuda.is_available():
+        torch.cuda.nvtx.range_push(f"stoke::{name}")
+        try:
+            yield
+        finally:
+            torch.cuda.nvtx.range_pop()
+    else:
+        yield
 
 
 class Stoke:
@@ -315,12 +334,12 @@ class Stoke:
     # --------------------------------------------------------------- hot loop
     def model(self, *args, **kwargs):
         """Forward call under the precision context."""
-        with self._runner.model_context:
+        with _phase("model"), self._runner.model_context:
             return self._model(*args, **kwargs)
 
     def loss(self, *args, **kwargs):
         """Loss call: computes, syncs for tracking, scales for accumulation."""
-        with self._runner.loss_context:
+        with _phase("loss"), self._runner.loss_context:
             if isinstance(self._loss, (list, tuple)):
                 loss = type(self._loss)(val(*args, **kwargs) for val in self._loss)
                 sync_loss = [self.detach_and_sync_loss(val) for val in loss]
@@ -365,13 +384,17 @@ class Stoke:
             if self._check_accum()
             else self._runner.grad_accum_context(self._model)
         )
-        with dist_cm:
+        with _phase("backward"), dist_cm:
             self._runner.backward_call(
                 loss=loss, model=self.model_access, optimizer=self._optimizer
             )
         self._backward_steps += 1
 
     def step(self):
+        with _phase("step"):
+            self._step_impl()
+
+    def _step_impl(self):
         if self._check_accum():
             if self._verbose and self.grad_accum > 0:
                 self.print(f"Gradient Accumulation Steps: {self.grad_accum}")

@@ -91,7 +91,9 @@ def test_save_load_gpu(tmp_path):
     assert s2.scaler.get_scale() == s.scaler.get_scale()
     x = torch.randn(4, 3, 32, 32, device="cuda")
     with torch.no_grad():
-        assert torch.allclose(s.model(x), s2.model(x))
+        # fp16 autocast + MIOpen algo selection is not bitwise deterministic
+        # across model instances; identical weights => fp16-eps-level agreement
+        assert torch.allclose(s.model(x), s2.model(x), rtol=1e-2, atol=5e-3)
 
 
 def test_native_ops_used_on_gpu():
