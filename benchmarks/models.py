@@ -21,24 +21,35 @@ import torch.nn.functional as F
 # ---------------------------------------------------------------------------
 # ResNet
 # ---------------------------------------------------------------------------
+from stoke.nn import FusedBNAct2d
+
+
+class _Downsample(nn.Module):
+    def __init__(self, cin, cout, stride):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, 1, stride, bias=False)
+        self.bn = FusedBNAct2d(cout, relu=False)
+
+    def forward(self, x):
+        return self.bn(self.conv(x))
+
+
 class BasicBlock(nn.Module):
     expansion = 1
 
     def __init__(self, cin, cout, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(cin, cout, 3, stride, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(cout)
+        self.bn1 = FusedBNAct2d(cout, relu=True)
         self.conv2 = nn.Conv2d(cout, cout, 3, 1, 1, bias=False)
-        self.bn2 = nn.BatchNorm2d(cout)
+        # bn2 fuses the residual add and final ReLU into one kernel pass
+        self.bn2 = FusedBNAct2d(cout, relu=True)
         self.downsample = downsample
 
     def forward(self, x):
-        idt = x
-        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
-        out = self.bn2(self.conv2(out))
-        if self.downsample is not None:
-            idt = self.downsample(x)
-        return F.relu(out + idt, inplace=True)
+        idt = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=idt)
 
 
 class Bottleneck(nn.Module):
@@ -48,21 +59,18 @@ class Bottleneck(nn.Module):
         super().__init__()
         cout = cmid * self.expansion
         self.conv1 = nn.Conv2d(cin, cmid, 1, 1, 0, bias=False)
-        self.bn1 = nn.BatchNorm2d(cmid)
+        self.bn1 = FusedBNAct2d(cmid, relu=True)
         self.conv2 = nn.Conv2d(cmid, cmid, 3, stride, 1, bias=False)
-        self.bn2 = nn.BatchNorm2d(cmid)
+        self.bn2 = FusedBNAct2d(cmid, relu=True)
         self.conv3 = nn.Conv2d(cmid, cout, 1, 1, 0, bias=False)
-        self.bn3 = nn.BatchNorm2d(cout)
+        self.bn3 = FusedBNAct2d(cout, relu=True)
         self.downsample = downsample
 
     def forward(self, x):
-        idt = x
-        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
-        out = F.relu(self.bn2(self.conv2(out)), inplace=True)
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            idt = self.downsample(x)
-        return F.relu(out + idt, inplace=True)
+        idt = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=idt)
 
 
 class ResNet(nn.Module):
@@ -72,14 +80,12 @@ class ResNet(nn.Module):
         if small_input:  # CIFAR-shape stem
             self.stem = nn.Sequential(
                 nn.Conv2d(3, 64, 3, 1, 1, bias=False),
-                nn.BatchNorm2d(64),
-                nn.ReLU(inplace=True),
+                FusedBNAct2d(64, relu=True),
             )
         else:
             self.stem = nn.Sequential(
                 nn.Conv2d(3, 64, 7, 2, 3, bias=False),
-                nn.BatchNorm2d(64),
-                nn.ReLU(inplace=True),
+                FusedBNAct2d(64, relu=True),
                 nn.MaxPool2d(3, 2, 1),
             )
         self.layer1 = self._make_layer(block, 64, layers[0], 1)
@@ -91,7 +97,7 @@ class ResNet(nn.Module):
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, (nn.BatchNorm2d, FusedBNAct2d)):
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
 
@@ -99,10 +105,7 @@ class ResNet(nn.Module):
         downsample = None
         cout = cmid * block.expansion
         if stride != 1 or self.cin != cout:
-            downsample = nn.Sequential(
-                nn.Conv2d(self.cin, cout, 1, stride, bias=False),
-                nn.BatchNorm2d(cout),
-            )
+            downsample = _Downsample(self.cin, cout, stride)
         layers = [block(self.cin, cmid, stride, downsample)]
         self.cin = cout
         for _ in range(1, n):

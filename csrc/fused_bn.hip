@@ -1,0 +1,434 @@
+// Fused NHWC BatchNorm (+ReLU, +residual) training kernels for MI355X.
+//
+// Motivation (profiles/resnet50_b256_steady_r01.txt): MIOpen's spatial BN
+// splits fwd into MeanVariance+Norm and bwd into DScaleDBias+DX, and eager
+// torch adds separate residual-add and ReLU kernels — together ~50% of a
+// ResNet-50 bf16 step on MI355X while convs are ~30%.  These kernels fuse:
+//   forward : one stats pass over x, one apply pass producing
+//             y = relu(scale*x + shift [+ residual])
+//   backward: one reduce pass (dgamma/dbeta with the ReLU mask folded in),
+//             one apply pass producing dx [and d_residual]
+// All passes are HBM-bound streaming over an [M, C] view (NHWC, channels
+// innermost): 16 B/lane vectorized bf16x8 access (guide Guideline 13),
+// fp32 accumulation, per-channel coefficients staged in LDS.
+//
+// Layout contract: x is channels-last (NHWC) contiguous, C % 8 == 0.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define BN_BLOCK 256
+
+namespace {
+
+typedef __hip_bfloat16 bf16;
+typedef float floatv4 __attribute__((ext_vector_type(4)));
+
+union bf16x8 {
+  uint4 u4;            // one 16-B load/store
+  unsigned short h[8];
+};
+
+__device__ __forceinline__ float b2f(unsigned short v) {
+  __hip_bfloat16_raw r;
+  r.x = v;
+  return __bfloat162float(*reinterpret_cast<bf16*>(&r));
+}
+
+__device__ __forceinline__ unsigned short f2b(float f) {
+  bf16 h = __float2bfloat16(f);
+  return *reinterpret_cast<unsigned short*>(&h);
+}
+
+// ---------------------------------------------------------------------------
+// Forward stats: per-channel sum and sum-of-squares over the M rows.
+// Each block owns a row-tile for ALL channels; threads own fixed channel
+// octets so partial sums live in registers; one LDS+atomic reduce per block.
+// Grid: (row_tiles, C/8 <= BN_BLOCK ? 1 : ceil(C/8/BN_BLOCK)) — in practice
+// C <= 2048 so C/8 <= 256 and one block spans all channels.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
+    const bf16x8* __restrict__ x, float* __restrict__ sum,
+    float* __restrict__ sqsum, long M, int c8, int rows_per_block) {
+  const int slots = c8;                      // channel octets per row
+  const int rows_par = BN_BLOCK / slots;     // rows processed concurrently
+  const int slot = threadIdx.x % slots;      // this thread's channel octet
+  const int rsub = threadIdx.x / slots;
+  long row0 = (long)blockIdx.x * rows_per_block;
+  long row_end = min(row0 + rows_per_block, M);
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float acc2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (rsub < rows_par) {
+    for (long r = row0 + rsub; r < row_end; r += rows_par) {
+      bf16x8 v = x[r * slots + slot];
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f = b2f(v.h[k]);
+        acc[k] += f;
+        acc2[k] += f * f;
+      }
+    }
+  }
+  // Reduce across the rows_par threads sharing a slot: LDS tree.
+  __shared__ float smem[BN_BLOCK * 8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = acc[k];
+  __syncthreads();
+  if (rsub == 0) {
+    #pragma unroll 1
+    for (int rr = 1; rr < rows_par; ++rr)
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) acc[k] += smem[(rr * slots + slot) * 8 + k];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k)
+      atomicAdd(&sum[slot * 8 + k], acc[k]);
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = acc2[k];
+  __syncthreads();
+  if (rsub == 0) {
+    #pragma unroll 1
+    for (int rr = 1; rr < rows_par; ++rr)
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) acc2[k] += smem[(rr * slots + slot) * 8 + k];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k)
+      atomicAdd(&sqsum[slot * 8 + k], acc2[k]);
+  }
+}
+
+// Finalize: mean/var -> apply coefficients + running-stat update (C threads).
+__global__ void bn_fwd_finalize_kernel(
+    const float* __restrict__ sum, const float* __restrict__ sqsum,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float* __restrict__ scale_out, float* __restrict__ shift_out,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    long M, int C, float eps, float momentum) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = sum[c] / (float)M;
+  float var = sqsum[c] / (float)M - m * m;
+  var = fmaxf(var, 0.f);
+  float invstd = rsqrtf(var + eps);
+  float sc = gamma[c] * invstd;
+  mean_out[c] = m;
+  invstd_out[c] = invstd;
+  scale_out[c] = sc;
+  shift_out[c] = beta[c] - m * sc;
+  if (running_mean != nullptr) {
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    float unbiased = var * (float)M / (float)max(M - 1, 1L);
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Forward apply: y = act(scale*x + shift [+ residual]).  Pure streaming.
+// ---------------------------------------------------------------------------
+template <bool RELU, bool RES>
+__global__ __launch_bounds__(BN_BLOCK) void bn_fwd_apply_kernel(
+    const bf16x8* __restrict__ x, const bf16x8* __restrict__ res,
+    bf16x8* __restrict__ y, const float* __restrict__ scale,
+    const float* __restrict__ shift, long n8, int c8) {
+  // stage per-channel coefficients in LDS (C <= 4096)
+  extern __shared__ float lds[];
+  float* s_scale = lds;
+  float* s_shift = lds + c8 * 8;
+  for (int i = threadIdx.x; i < c8 * 8; i += BN_BLOCK) {
+    s_scale[i] = scale[i];
+    s_shift[i] = shift[i];
+  }
+  __syncthreads();
+  long stride = (long)gridDim.x * BN_BLOCK;
+  for (long i = (long)blockIdx.x * BN_BLOCK + threadIdx.x; i < n8; i += stride) {
+    int slot = (int)(i % c8) * 8;
+    bf16x8 v = x[i];
+    bf16x8 r;
+    if (RES) r = res[i];
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f = b2f(v.h[k]) * s_scale[slot + k] + s_shift[slot + k];
+      if (RES) f += b2f(r.h[k]);
+      if (RELU) f = fmaxf(f, 0.f);
+      o.h[k] = f2b(f);
+    }
+    y[i] = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward reduce: dbeta = sum(dy_eff), dgamma = sum(dy_eff * xhat) with the
+// ReLU mask (y > 0) folded in; optionally writes d_residual = dy_eff.
+// Same block geometry as the stats kernel.
+// ---------------------------------------------------------------------------
+template <bool RELU, bool RES>
+__global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
+    const bf16x8* __restrict__ x, const bf16x8* __restrict__ dy,
+    const bf16x8* __restrict__ y, bf16x8* __restrict__ dres,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ dbeta, float* __restrict__ dgamma, long M, int c8,
+    int rows_per_block) {
+  const int slots = c8;
+  const int rows_par = BN_BLOCK / slots;
+  const int slot = threadIdx.x % slots;
+  const int rsub = threadIdx.x / slots;
+  long row0 = (long)blockIdx.x * rows_per_block;
+  long row_end = min(row0 + rows_per_block, M);
+  float m[8], is[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    m[k] = mean[slot * 8 + k];
+    is[k] = invstd[slot * 8 + k];
+  }
+  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (rsub < rows_par) {
+    for (long r = row0 + rsub; r < row_end; r += rows_par) {
+      long i = r * slots + slot;
+      bf16x8 vdy = dy[i];
+      bf16x8 vx = x[i];
+      bf16x8 vy;
+      if (RELU) vy = y[i];
+      bf16x8 vdr;
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = b2f(vdy.h[k]);
+        if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+        if (RES) vdr.h[k] = f2b(g);
+        float xh = (b2f(vx.h[k]) - m[k]) * is[k];
+        db[k] += g;
+        dg[k] += g * xh;
+      }
+      if (RES) dres[i] = vdr;
+    }
+  }
+  __shared__ float smem[BN_BLOCK * 8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = db[k];
+  __syncthreads();
+  if (rsub == 0) {
+    #pragma unroll 1
+    for (int rr = 1; rr < rows_par; ++rr)
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) db[k] += smem[(rr * slots + slot) * 8 + k];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) atomicAdd(&dbeta[slot * 8 + k], db[k]);
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = dg[k];
+  __syncthreads();
+  if (rsub == 0) {
+    #pragma unroll 1
+    for (int rr = 1; rr < rows_par; ++rr)
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) dg[k] += smem[(rr * slots + slot) * 8 + k];
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) atomicAdd(&dgamma[slot * 8 + k], dg[k]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward apply: dx = (gamma*invstd) * (dy_eff - dbeta/M - xhat*dgamma/M)
+// ---------------------------------------------------------------------------
+template <bool RELU>
+__global__ __launch_bounds__(BN_BLOCK) void bn_bwd_apply_kernel(
+    const bf16x8* __restrict__ x, const bf16x8* __restrict__ dy,
+    const bf16x8* __restrict__ y, bf16x8* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ dbeta,
+    const float* __restrict__ dgamma, long n8, int c8, float invM) {
+  extern __shared__ float lds[];
+  float* s_a = lds;               // gamma*invstd
+  float* s_m = lds + c8 * 8;      // mean
+  float* s_is = lds + c8 * 16;    // invstd
+  float* s_db = lds + c8 * 24;    // dbeta/M
+  float* s_dg = lds + c8 * 32;    // dgamma/M
+  for (int i = threadIdx.x; i < c8 * 8; i += BN_BLOCK) {
+    s_a[i] = gamma[i] * invstd[i];
+    s_m[i] = mean[i];
+    s_is[i] = invstd[i];
+    s_db[i] = dbeta[i] * invM;
+    s_dg[i] = dgamma[i] * invM;
+  }
+  __syncthreads();
+  long stride = (long)gridDim.x * BN_BLOCK;
+  for (long i = (long)blockIdx.x * BN_BLOCK + threadIdx.x; i < n8; i += stride) {
+    int slot = (int)(i % c8) * 8;
+    bf16x8 vdy = dy[i];
+    bf16x8 vx = x[i];
+    bf16x8 vy;
+    if (RELU) vy = y[i];
+    bf16x8 o;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float g = b2f(vdy.h[k]);
+      if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
+      float xh = (b2f(vx.h[k]) - s_m[slot + k]) * s_is[slot + k];
+      float d = s_a[slot + k] * (g - s_db[slot + k] - xh * s_dg[slot + k]);
+      o.h[k] = f2b(d);
+    }
+    dx[i] = o;
+  }
+}
+
+int pick_row_tiles(long M, int c8) {
+  // enough blocks to fill 256 CUs x ~4 blocks, but cap tiny inputs
+  long target_blocks = 2048;
+  long rows_per_block = (M + target_blocks - 1) / target_blocks;
+  if (rows_per_block < 16) rows_per_block = 16;
+  return (int)rows_per_block;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Entry points.  x: [M, C] view of NHWC bf16; params fp32.
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> bn_fwd_train(
+    at::Tensor x, c10::optional<at::Tensor> residual, at::Tensor gamma,
+    at::Tensor beta, c10::optional<at::Tensor> running_mean,
+    c10::optional<at::Tensor> running_var, double eps, double momentum,
+    bool relu) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "bn_fwd_train: bf16 only");
+  const long C = x.size(-1);
+  TORCH_CHECK(C % 8 == 0 && C <= 4096, "bn: C must be %8==0 and <=4096");
+  const long M = x.numel() / C;
+  const int c8 = (int)(C / 8);
+  TORCH_CHECK(BN_BLOCK % c8 == 0 || c8 >= BN_BLOCK,
+              "bn: C/8 must divide 256 or exceed it");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto opts = gamma.options();
+  auto sum = at::zeros({C}, opts);
+  auto sqsum = at::zeros({C}, opts);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto scale = at::empty({C}, opts);
+  auto shift = at::empty({C}, opts);
+  auto y = at::empty_like(x);
+  const int slots = c8 <= BN_BLOCK ? c8 : BN_BLOCK;
+  TORCH_CHECK(slots == c8, "bn: C too large for single-block-span layout");
+  int rpb = pick_row_tiles(M, c8);
+  int nblocks = (int)((M + rpb - 1) / rpb);
+  hipLaunchKernelGGL(bn_fwd_stats_kernel, dim3(nblocks), dim3(BN_BLOCK), 0,
+                     stream, (const bf16x8*)x.data_ptr(),
+                     sum.data_ptr<float>(), sqsum.data_ptr<float>(), M, c8,
+                     rpb);
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                     0, stream, sum.data_ptr<float>(), sqsum.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),
+                     running_mean ? running_mean->data_ptr<float>() : nullptr,
+                     running_var ? running_var->data_ptr<float>() : nullptr,
+                     M, (int)C, (float)eps, (float)momentum);
+  long n8 = M * c8;
+  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
+  size_t lds_bytes = (size_t)C * 2 * sizeof(float);
+  const bf16x8* resp =
+      residual ? (const bf16x8*)residual->data_ptr() : nullptr;
+  auto launch_apply = [&](auto relu_t, auto res_t) {
+    hipLaunchKernelGGL((bn_fwd_apply_kernel<decltype(relu_t)::value,
+                                            decltype(res_t)::value>),
+                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       (const bf16x8*)x.data_ptr(), resp,
+                       (bf16x8*)y.data_ptr(), scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), n8, c8);
+  };
+  if (relu && residual) launch_apply(std::true_type{}, std::true_type{});
+  else if (relu) launch_apply(std::true_type{}, std::false_type{});
+  else if (residual) launch_apply(std::false_type{}, std::true_type{});
+  else launch_apply(std::false_type{}, std::false_type{});
+  return {y, mean, invstd};
+}
+
+at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> residual,
+                       at::Tensor gamma, at::Tensor beta,
+                       at::Tensor running_mean, at::Tensor running_var,
+                       double eps, bool relu) {
+  const long C = x.size(-1);
+  const long M = x.numel() / C;
+  const int c8 = (int)(C / 8);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto invstd = (running_var + eps).rsqrt();
+  auto scale = (gamma * invstd).contiguous();
+  auto shift = (beta - running_mean * scale).contiguous();
+  auto y = at::empty_like(x);
+  long n8 = M * c8;
+  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
+  size_t lds_bytes = (size_t)C * 2 * sizeof(float);
+  const bf16x8* resp =
+      residual ? (const bf16x8*)residual->data_ptr() : nullptr;
+  auto launch_apply = [&](auto relu_t, auto res_t) {
+    hipLaunchKernelGGL((bn_fwd_apply_kernel<decltype(relu_t)::value,
+                                            decltype(res_t)::value>),
+                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       (const bf16x8*)x.data_ptr(), resp,
+                       (bf16x8*)y.data_ptr(), scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), n8, c8);
+  };
+  if (relu && residual) launch_apply(std::true_type{}, std::true_type{});
+  else if (relu) launch_apply(std::true_type{}, std::false_type{});
+  else if (residual) launch_apply(std::false_type{}, std::true_type{});
+  else launch_apply(std::false_type{}, std::false_type{});
+  return y;
+}
+
+std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                               at::Tensor mean, at::Tensor invstd,
+                               at::Tensor gamma, bool relu, bool needs_dres) {
+  const long C = x.size(-1);
+  const long M = x.numel() / C;
+  const int c8 = (int)(C / 8);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto opts = gamma.options();
+  auto dbeta = at::zeros({C}, opts);
+  auto dgamma = at::zeros({C}, opts);
+  auto dx = at::empty_like(x);
+  at::Tensor dres;
+  bf16x8* dresp = nullptr;
+  if (needs_dres) {
+    dres = at::empty_like(x);
+    dresp = (bf16x8*)dres.data_ptr();
+  }
+  int rpb = pick_row_tiles(M, c8);
+  int nblocks = (int)((M + rpb - 1) / rpb);
+  auto launch_reduce = [&](auto relu_t, auto res_t) {
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<decltype(relu_t)::value,
+                                             decltype(res_t)::value>),
+                       dim3(nblocks), dim3(BN_BLOCK), 0, stream,
+                       (const bf16x8*)x.data_ptr(),
+                       (const bf16x8*)dy.data_ptr(),
+                       (const bf16x8*)y.data_ptr(), dresp,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), M,
+                       c8, rpb);
+  };
+  if (relu && needs_dres) launch_reduce(std::true_type{}, std::true_type{});
+  else if (relu) launch_reduce(std::true_type{}, std::false_type{});
+  else if (needs_dres) launch_reduce(std::false_type{}, std::true_type{});
+  else launch_reduce(std::false_type{}, std::false_type{});
+  long n8 = M * c8;
+  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
+  size_t lds_bytes = (size_t)C * 5 * sizeof(float);
+  auto launch_apply = [&](auto relu_t) {
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<decltype(relu_t)::value>),
+                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       (const bf16x8*)x.data_ptr(),
+                       (const bf16x8*)dy.data_ptr(),
+                       (const bf16x8*)y.data_ptr(), (bf16x8*)dx.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                       dgamma.data_ptr<float>(), n8, c8, 1.f / (float)M);
+  };
+  if (relu) launch_apply(std::true_type{});
+  else launch_apply(std::false_type{});
+  if (!needs_dres) dres = at::Tensor();
+  return {dx, dgamma, dbeta, dres};
+}

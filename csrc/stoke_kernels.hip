@@ -400,7 +400,26 @@ void amp_update_scale_(at::Tensor scale, at::Tensor growth_tracker,
                      (float)backoff_factor, (int)growth_interval);
 }
 
+// Fused NHWC BatchNorm kernels (csrc/fused_bn.hip)
+std::vector<at::Tensor> bn_fwd_train(
+    at::Tensor x, c10::optional<at::Tensor> residual, at::Tensor gamma,
+    at::Tensor beta, c10::optional<at::Tensor> running_mean,
+    c10::optional<at::Tensor> running_var, double eps, double momentum,
+    bool relu);
+at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> residual,
+                       at::Tensor gamma, at::Tensor beta,
+                       at::Tensor running_mean, at::Tensor running_var,
+                       double eps, bool relu);
+std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                               at::Tensor mean, at::Tensor invstd,
+                               at::Tensor gamma, bool relu, bool needs_dres);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_train", &bn_fwd_train,
+        "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
+  m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN forward (eval)");
+  m.def("bn_bwd", &bn_bwd,
+        "fused NHWC bf16 BN backward: reduce(+relu mask) + dx(+dres)");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

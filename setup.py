@@ -31,7 +31,7 @@ setup(
     ext_modules=[
         cpp_extension.CUDAExtension(
             name="stoke._C",
-            sources=["csrc/stoke_kernels.hip"],
+            sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

@@ -26,6 +26,10 @@ class FusedAdamW(torch.optim.Optimizer):
     # StokeGradScaler passes found_inf to step() so the skip-on-overflow
     # decision stays on device (no host sync).
     step_supports_found_inf = True
+    # Multi-tensor-apply re-collects grads each step, so set_to_none zeroing
+    # is safe and skips the fillBuffer pass classic fused optimizers need
+    # (1.7% of a ResNet-50 step in profiles/resnet50_b256_steady_r01.txt).
+    zero_grad_prefers_none = True
 
     def __init__(
         self,

@@ -72,7 +72,10 @@ def zero_optimizer_grads(
     are zeroed in place; everything else gets ``set_to_none=True`` (reference
     semantics, ``utils.py:83-106``).
     """
-    if (optimizer.__class__.__name__.find("Fused") == -1) and not apex and not horovod:
+    prefers_none = getattr(optimizer, "zero_grad_prefers_none", False)
+    if prefers_none or (
+        optimizer.__class__.__name__.find("Fused") == -1 and not apex and not horovod
+    ):
         optimizer.zero_grad(set_to_none=True)
     else:
         optimizer.zero_grad(set_to_none=False)
