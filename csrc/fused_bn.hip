@@ -46,13 +46,18 @@ __device__ __forceinline__ unsigned short f2b(float f) {
 // ---------------------------------------------------------------------------
 // Forward stats: per-channel sum and sum-of-squares over the M rows.
 // Each block owns a row-tile for ALL channels; threads own fixed channel
-// octets so partial sums live in registers; one LDS+atomic reduce per block.
-// Grid: (row_tiles, C/8 <= BN_BLOCK ? 1 : ceil(C/8/BN_BLOCK)) — in practice
-// C <= 2048 so C/8 <= 256 and one block spans all channels.
+// octets so partial sums live in registers; one LDS reduce per block, then
+// the block writes its per-channel partials to ``part`` ([nblocks, 2C]:
+// sums first, then sum-of-squares).  NO global atomics: ~2000 blocks
+// atomically RMW-ing the same C addresses serialize per address (~measured
+// 400+us/launch on the 14x14 layers); per-block partials + a coalesced
+// second-stage reduce in the finalize kernel run at streaming speed.
+// Grid is capped at BN_NBLK blocks (512 x 4 waves = 8 waves/CU on 256 CUs,
+// enough to saturate HBM for a pure streaming kernel).
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
-    const bf16x8* __restrict__ x, float* __restrict__ sum,
-    float* __restrict__ sqsum, long M, int c8, int rows_per_block) {
+    const bf16x8* __restrict__ x, float* __restrict__ part, long M, int c8,
+    int rows_per_block) {
   const int slots = c8;                      // channel octets per row
   const int rows_par = BN_BLOCK / slots;     // rows processed concurrently
   const int slot = threadIdx.x % slots;      // this thread's channel octet
@@ -74,6 +79,8 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
   }
   // Reduce across the rows_par threads sharing a slot: LDS tree.
   __shared__ float smem[BN_BLOCK * 8];
+  const int C = c8 * 8;
+  float* out = part + (long)blockIdx.x * 2 * C;
   #pragma unroll
   for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = acc[k];
   __syncthreads();
@@ -83,8 +90,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
       #pragma unroll
       for (int k = 0; k < 8; ++k) acc[k] += smem[(rr * slots + slot) * 8 + k];
     #pragma unroll
-    for (int k = 0; k < 8; ++k)
-      atomicAdd(&sum[slot * 8 + k], acc[k]);
+    for (int k = 0; k < 8; ++k) out[slot * 8 + k] = acc[k];
   }
   __syncthreads();
   #pragma unroll
@@ -96,14 +102,16 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
       #pragma unroll
       for (int k = 0; k < 8; ++k) acc2[k] += smem[(rr * slots + slot) * 8 + k];
     #pragma unroll
-    for (int k = 0; k < 8; ++k)
-      atomicAdd(&sqsum[slot * 8 + k], acc2[k]);
+    for (int k = 0; k < 8; ++k) out[C + slot * 8 + k] = acc2[k];
   }
 }
 
-// Finalize: mean/var -> apply coefficients + running-stat update (C threads).
+// Finalize: reduce block partials, mean/var -> apply coefficients +
+// running-stat update.  One thread per channel; the partial loop reads
+// [nblocks, 2C] with consecutive threads on consecutive channels, so every
+// iteration is one coalesced row read.
 __global__ void bn_fwd_finalize_kernel(
-    const float* __restrict__ sum, const float* __restrict__ sqsum,
+    const float* __restrict__ part, int nblocks,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
     float* __restrict__ scale_out, float* __restrict__ shift_out,
@@ -111,8 +119,14 @@ __global__ void bn_fwd_finalize_kernel(
     long M, int C, float eps, float momentum) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = sum[c] / (float)M;
-  float var = sqsum[c] / (float)M - m * m;
+  float s = 0.f, s2 = 0.f;
+  #pragma unroll 4
+  for (int p = 0; p < nblocks; ++p) {
+    s += part[(long)p * 2 * C + c];
+    s2 += part[(long)p * 2 * C + C + c];
+  }
+  float m = s / (float)M;
+  float var = s2 / (float)M - m * m;
   var = fmaxf(var, 0.f);
   float invstd = rsqrtf(var + eps);
   float sc = gamma[c] * invstd;
@@ -172,8 +186,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
     const bf16x8* __restrict__ x, const bf16x8* __restrict__ dy,
     const bf16x8* __restrict__ y, bf16x8* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ dbeta, float* __restrict__ dgamma, long M, int c8,
-    int rows_per_block) {
+    float* __restrict__ part, long M, int c8, int rows_per_block) {
   const int slots = c8;
   const int rows_par = BN_BLOCK / slots;
   const int slot = threadIdx.x % slots;
@@ -209,6 +222,8 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
     }
   }
   __shared__ float smem[BN_BLOCK * 8];
+  const int C = c8 * 8;
+  float* out = part + (long)blockIdx.x * 2 * C;
   #pragma unroll
   for (int k = 0; k < 8; ++k) smem[threadIdx.x * 8 + k] = db[k];
   __syncthreads();
@@ -218,7 +233,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
       #pragma unroll
       for (int k = 0; k < 8; ++k) db[k] += smem[(rr * slots + slot) * 8 + k];
     #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&dbeta[slot * 8 + k], db[k]);
+    for (int k = 0; k < 8; ++k) out[slot * 8 + k] = db[k];
   }
   __syncthreads();
   #pragma unroll
@@ -230,8 +245,24 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
       #pragma unroll
       for (int k = 0; k < 8; ++k) dg[k] += smem[(rr * slots + slot) * 8 + k];
     #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&dgamma[slot * 8 + k], dg[k]);
+    for (int k = 0; k < 8; ++k) out[C + slot * 8 + k] = dg[k];
   }
+}
+
+// Reduce bwd partials ([nblocks, 2C]: dbeta rows then dgamma) -> dbeta, dgamma.
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
+                                       int nblocks, float* __restrict__ dbeta,
+                                       float* __restrict__ dgamma, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float db = 0.f, dg = 0.f;
+  #pragma unroll 4
+  for (int p = 0; p < nblocks; ++p) {
+    db += part[(long)p * 2 * C + c];
+    dg += part[(long)p * 2 * C + C + c];
+  }
+  dbeta[c] = db;
+  dgamma[c] = dg;
 }
 
 // ---------------------------------------------------------------------------
@@ -279,8 +310,10 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_apply_kernel(
 }
 
 int pick_row_tiles(long M, int c8) {
-  // enough blocks to fill 256 CUs x ~4 blocks, but cap tiny inputs
-  long target_blocks = 2048;
+  // BN_NBLK blocks of 4 waves = 8 waves/CU on 256 CUs: saturates HBM for a
+  // streaming reduction while keeping the partial buffer / second-stage
+  // reduce small.  More blocks would only lengthen the finalize loop.
+  long target_blocks = 512;
   long rows_per_block = (M + target_blocks - 1) / target_blocks;
   if (rows_per_block < 16) rows_per_block = 16;
   return (int)rows_per_block;
@@ -305,8 +338,6 @@ std::vector<at::Tensor> bn_fwd_train(
               "bn: C/8 must divide 256 or exceed it");
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto opts = gamma.options();
-  auto sum = at::zeros({C}, opts);
-  auto sqsum = at::zeros({C}, opts);
   auto mean = at::empty({C}, opts);
   auto invstd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts);
@@ -316,12 +347,12 @@ std::vector<at::Tensor> bn_fwd_train(
   TORCH_CHECK(slots == c8, "bn: C too large for single-block-span layout");
   int rpb = pick_row_tiles(M, c8);
   int nblocks = (int)((M + rpb - 1) / rpb);
+  auto part = at::empty({(long)nblocks * 2 * C}, opts);
   hipLaunchKernelGGL(bn_fwd_stats_kernel, dim3(nblocks), dim3(BN_BLOCK), 0,
                      stream, (const bf16x8*)x.data_ptr(),
-                     sum.data_ptr<float>(), sqsum.data_ptr<float>(), M, c8,
-                     rpb);
+                     part.data_ptr<float>(), M, c8, rpb);
   hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                     0, stream, sum.data_ptr<float>(), sqsum.data_ptr<float>(),
+                     0, stream, part.data_ptr<float>(), nblocks,
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      scale.data_ptr<float>(), shift.data_ptr<float>(),
@@ -388,8 +419,8 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   const int c8 = (int)(C / 8);
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto opts = gamma.options();
-  auto dbeta = at::zeros({C}, opts);
-  auto dgamma = at::zeros({C}, opts);
+  auto dbeta = at::empty({C}, opts);
+  auto dgamma = at::empty({C}, opts);
   auto dx = at::empty_like(x);
   at::Tensor dres;
   bf16x8* dresp = nullptr;
@@ -399,6 +430,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   }
   int rpb = pick_row_tiles(M, c8);
   int nblocks = (int)((M + rpb - 1) / rpb);
+  auto part = at::empty({(long)nblocks * 2 * C}, opts);
   auto launch_reduce = [&](auto relu_t, auto res_t) {
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<decltype(relu_t)::value,
                                              decltype(res_t)::value>),
@@ -407,13 +439,16 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                        (const bf16x8*)dy.data_ptr(),
                        (const bf16x8*)y.data_ptr(), dresp,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), M,
-                       c8, rpb);
+                       part.data_ptr<float>(), M, c8, rpb);
   };
   if (relu && needs_dres) launch_reduce(std::true_type{}, std::true_type{});
   else if (relu) launch_reduce(std::true_type{}, std::false_type{});
   else if (needs_dres) launch_reduce(std::false_type{}, std::true_type{});
   else launch_reduce(std::false_type{}, std::false_type{});
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                     0, stream, part.data_ptr<float>(), nblocks,
+                     dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     (int)C);
   long n8 = M * c8;
   int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
   size_t lds_bytes = (size_t)C * 5 * sizeof(float);

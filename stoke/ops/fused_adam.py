@@ -39,13 +39,22 @@ class FusedAdamW(torch.optim.Optimizer):
         eps: float = 1e-8,
         weight_decay: float = 1e-2,
         adam_w_mode: bool = True,
+        offload_state: bool = False,
     ):
+        """``offload_state=True`` keeps exp_avg/exp_avg_sq (and the fp32
+        master for bf16 params) in pinned host memory — the in-house
+        equivalent of DeepSpeed's CPU optimizer offload
+        (``DeepspeedOffloadOptimizerConfig``, reference ``configs.py:308-342``).
+        Each step streams state H2D, runs the fused HIP kernel, and streams
+        it back D2H (pinned + non_blocking, ordered on the current stream);
+        HBM then only holds params+grads, trading step time for capacity."""
         if lr < 0.0:
             raise ValueError(f"Invalid learning rate: {lr}")
         defaults = dict(
             lr=lr, betas=betas, eps=eps, weight_decay=weight_decay,
             adam_w_mode=adam_w_mode,
         )
+        self.offload_state = offload_state
         super().__init__(params, defaults)
 
     @torch.no_grad()
@@ -59,6 +68,8 @@ class FusedAdamW(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        if self.offload_state:
+            return self._step_offload(loss, found_inf, inv_scale)
         for group in self.param_groups:
             beta1, beta2 = group["betas"]
             wd = group["weight_decay"] if group["adam_w_mode"] else 0.0
@@ -102,4 +113,60 @@ class FusedAdamW(torch.optim.Optimizer):
                     group["lr"], beta1, beta2, group["eps"], wd,
                     found_inf=found_inf, inv_scale=inv_scale, masters=bf16_w,
                 )
+        return loss
+
+    def _step_offload(self, loss, found_inf, inv_scale):
+        """Pinned-host state variant: per-parameter H2D -> fused kernel -> D2H.
+
+        All copies are non_blocking on the current stream, so kernel K of
+        parameter i overlaps the H2D of parameter i+1 at the copy-engine
+        level while strict stream ordering keeps the math correct.
+        """
+        on_gpu = any(
+            p.is_cuda for g in self.param_groups for p in g["params"]
+        )
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            wd = group["weight_decay"] if group["adam_w_mode"] else 0.0
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros(
+                        p.shape, dtype=torch.float32, device="cpu",
+                        pin_memory=on_gpu,
+                    )
+                    state["exp_avg_sq"] = torch.zeros(
+                        p.shape, dtype=torch.float32, device="cpu",
+                        pin_memory=on_gpu,
+                    )
+                    if p.dtype == torch.bfloat16:
+                        m = p.detach().float().cpu()
+                        state["master"] = m.pin_memory() if on_gpu else m
+                state["step"] += 1
+                m_dev = state["exp_avg"].to(p.device, non_blocking=True)
+                v_dev = state["exp_avg_sq"].to(p.device, non_blocking=True)
+                if p.dtype == torch.bfloat16:
+                    w_dev = state["master"].to(p.device, non_blocking=True)
+                    ops.fused_adamw_(
+                        [p.data], [p.grad.data], [m_dev], [v_dev],
+                        state["step"], group["lr"], beta1, beta2,
+                        group["eps"], wd, found_inf=found_inf,
+                        inv_scale=inv_scale, masters=[w_dev],
+                    )
+                    state["master"].copy_(w_dev, non_blocking=True)
+                else:
+                    ops.fused_adamw_(
+                        [p.data], [p.grad.data], [m_dev], [v_dev],
+                        state["step"], group["lr"], beta1, beta2,
+                        group["eps"], wd, found_inf=found_inf,
+                        inv_scale=inv_scale,
+                    )
+                state["exp_avg"].copy_(m_dev, non_blocking=True)
+                state["exp_avg_sq"].copy_(v_dev, non_blocking=True)
+        # Host buffers must not be reused before their D2H copies land.
+        if on_gpu:
+            torch.cuda.synchronize()
         return loss

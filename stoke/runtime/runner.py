@@ -186,6 +186,20 @@ class StokeRunner:
             convert_bn = True
         if convert_bn:
             model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+        # In-house activation checkpointing (the reference delegated this to
+        # the DeepSpeed engine via DeepspeedActivationCheckpointingConfig)
+        if (
+            s.is_distributed_deepspeed
+            and s.deepspeed_config.activation_checkpointing is not None
+        ):
+            from stoke.nn import apply_activation_checkpointing
+
+            n = apply_activation_checkpointing(model)
+            if self._verbose:
+                self.print_device(
+                    f"Activation checkpointing: wrapped {n} module(s)",
+                    rank=self._info_rank,
+                )
         if self._shard == "fsdp":
             fcfg = s.fsdp_config
             wrapped = StokeFSDPModule(
@@ -246,6 +260,30 @@ class StokeRunner:
 
     def build_optimizer(self, optimizer, optimizer_kwargs, model):
         """Instantiate the optimizer (plain or OSS-sharded)."""
+        s = self._status
+        if (
+            s.is_distributed_deepspeed
+            and s.deepspeed_config.zero_optimization is not None
+            and s.deepspeed_config.zero_optimization.offload_optimizer is not None
+        ):
+            off = s.deepspeed_config.zero_optimization.offload_optimizer
+            dev = getattr(off.device, "value", off.device)
+            if dev == "cpu":
+                import inspect
+
+                if "offload_state" in inspect.signature(optimizer).parameters:
+                    optimizer_kwargs = {**optimizer_kwargs, "offload_state": True}
+                else:
+                    raise ValueError(
+                        "Stoke -- CPU optimizer-state offload requires an "
+                        "optimizer supporting offload_state (e.g. stoke "
+                        "FusedAdamW); got "
+                        f"{getattr(optimizer, '__name__', optimizer)}"
+                    )
+            elif dev == "nvme":
+                raise ValueError(
+                    "Stoke -- NVMe optimizer offload is not implemented"
+                )
         if self._shard in ("oss", "sddp"):
             ocfg = self._status.oss_config
             opt = OSSOptimizer(
