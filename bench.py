@@ -110,6 +110,13 @@ def main():
             return torch.nn.functional.cross_entropy(
                 logits.reshape(-1, logits.shape[-1]).float(), target.reshape(-1)
             )
+        # Pure-bf16 weights for the non-FSDP LM paths: FusedAdamW keeps fp32
+        # masters (HIP bf16 kernel), GEMMs skip the autocast weight casts,
+        # and DDP/OSS collectives move half the bytes.  The FSDP engine owns
+        # its own fp32 flat shards and bf16 compute copies, so it takes the
+        # fp32 module.
+        if not (args.model == "llama-fsdp" and distributed):
+            model = model.bfloat16()
     else:
         loss_fn = torch.nn.CrossEntropyLoss()
 

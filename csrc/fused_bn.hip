@@ -107,24 +107,53 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
 }
 
 // Finalize: reduce block partials, mean/var -> apply coefficients +
-// running-stat update.  One thread per channel; the partial loop reads
-// [nblocks, 2C] with consecutive threads on consecutive channels, so every
-// iteration is one coalesced row read.
-__global__ void bn_fwd_finalize_kernel(
+// running-stat update.  1024 threads laid out as 16 partial-lanes x 64
+// channel-lanes per block: each wave reads 64 consecutive channels of one
+// partial row (fully coalesced) and 16 rows are in flight, so the
+// [nblocks, 2C] sweep is latency-hidden even when C is small (a
+// one-thread-per-channel version measured 42 us/call — 4.4 ms/step —
+// because 1-8 blocks of serial loads ran on a handful of CUs).
+#define FIN_PPAR 16
+#define FIN_CH 64
+
+__device__ __forceinline__ void finalize_reduce_pair(
+    const float* __restrict__ part, int nblocks, int C, int c, bool valid,
+    float* s_out, float* s2_out) {
+  const int pp = threadIdx.x / FIN_CH;   // partial lane
+  const int cc = threadIdx.x % FIN_CH;
+  float s = 0.f, s2 = 0.f;
+  if (valid) {
+    for (int p = pp; p < nblocks; p += FIN_PPAR) {
+      s += part[(long)p * 2 * C + c];
+      s2 += part[(long)p * 2 * C + C + c];
+    }
+  }
+  __shared__ float smem[2 * FIN_PPAR * FIN_CH];
+  smem[pp * FIN_CH + cc] = s;
+  smem[FIN_PPAR * FIN_CH + pp * FIN_CH + cc] = s2;
+  __syncthreads();
+  if (pp == 0) {
+    #pragma unroll
+    for (int r = 1; r < FIN_PPAR; ++r) {
+      s += smem[r * FIN_CH + cc];
+      s2 += smem[FIN_PPAR * FIN_CH + r * FIN_CH + cc];
+    }
+    *s_out = s;
+    *s2_out = s2;
+  }
+}
+
+__global__ __launch_bounds__(FIN_PPAR * FIN_CH) void bn_fwd_finalize_kernel(
     const float* __restrict__ part, int nblocks,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
     float* __restrict__ scale_out, float* __restrict__ shift_out,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     long M, int C, float eps, float momentum) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float s = 0.f, s2 = 0.f;
-  #pragma unroll 4
-  for (int p = 0; p < nblocks; ++p) {
-    s += part[(long)p * 2 * C + c];
-    s2 += part[(long)p * 2 * C + C + c];
-  }
+  const int c = blockIdx.x * FIN_CH + threadIdx.x % FIN_CH;
+  float s, s2;
+  finalize_reduce_pair(part, nblocks, C, c, c < C, &s, &s2);
+  if (threadIdx.x >= FIN_CH || c >= C) return;
   float m = s / (float)M;
   float var = s2 / (float)M - m * m;
   var = fmaxf(var, 0.f);
@@ -148,27 +177,34 @@ template <bool RELU, bool RES>
 __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_apply_kernel(
     const bf16x8* __restrict__ x, const bf16x8* __restrict__ res,
     bf16x8* __restrict__ y, const float* __restrict__ scale,
-    const float* __restrict__ shift, long n8, int c8) {
-  // stage per-channel coefficients in LDS (C <= 4096)
-  extern __shared__ float lds[];
-  float* s_scale = lds;
-  float* s_shift = lds + c8 * 8;
-  for (int i = threadIdx.x; i < c8 * 8; i += BN_BLOCK) {
-    s_scale[i] = scale[i];
-    s_shift[i] = shift[i];
+    const float* __restrict__ shift, long M, int c8, int rows_per_block) {
+  // Row-tile geometry (same as the stats kernel): each thread owns a fixed
+  // channel octet, so its scale/shift coefficients live in 16 registers and
+  // there is no per-element modulo; a wave still covers 64 consecutive
+  // octets of one row pair = fully coalesced 16 B/lane streaming.
+  const int slots = c8;
+  const int rows_par = BN_BLOCK / slots;
+  const int slot = threadIdx.x % slots;
+  const int rsub = threadIdx.x / slots;
+  float sc[8], sh[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    sc[k] = scale[slot * 8 + k];
+    sh[k] = shift[slot * 8 + k];
   }
-  __syncthreads();
-  long stride = (long)gridDim.x * BN_BLOCK;
-  for (long i = (long)blockIdx.x * BN_BLOCK + threadIdx.x; i < n8; i += stride) {
-    int slot = (int)(i % c8) * 8;
+  long row0 = (long)blockIdx.x * rows_per_block;
+  long row_end = min(row0 + rows_per_block, M);
+  if (rsub >= rows_par) return;
+  for (long r = row0 + rsub; r < row_end; r += rows_par) {
+    long i = r * slots + slot;
     bf16x8 v = x[i];
-    bf16x8 r;
-    if (RES) r = res[i];
+    bf16x8 rr;
+    if (RES) rr = res[i];
     bf16x8 o;
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      float f = b2f(v.h[k]) * s_scale[slot + k] + s_shift[slot + k];
-      if (RES) f += b2f(r.h[k]);
+      float f = b2f(v.h[k]) * sc[k] + sh[k];
+      if (RES) f += b2f(rr.h[k]);
       if (RELU) f = fmaxf(f, 0.f);
       o.h[k] = f2b(f);
     }
@@ -249,18 +285,15 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
   }
 }
 
-// Reduce bwd partials ([nblocks, 2C]: dbeta rows then dgamma) -> dbeta, dgamma.
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
-                                       int nblocks, float* __restrict__ dbeta,
-                                       float* __restrict__ dgamma, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float db = 0.f, dg = 0.f;
-  #pragma unroll 4
-  for (int p = 0; p < nblocks; ++p) {
-    db += part[(long)p * 2 * C + c];
-    dg += part[(long)p * 2 * C + C + c];
-  }
+// Reduce bwd partials ([nblocks, 2C]: dbeta rows then dgamma) -> dbeta,
+// dgamma.  Same 16x64 geometry as bn_fwd_finalize_kernel.
+__global__ __launch_bounds__(FIN_PPAR * FIN_CH) void bn_bwd_finalize_kernel(
+    const float* __restrict__ part, int nblocks, float* __restrict__ dbeta,
+    float* __restrict__ dgamma, int C) {
+  const int c = blockIdx.x * FIN_CH + threadIdx.x % FIN_CH;
+  float db, dg;
+  finalize_reduce_pair(part, nblocks, C, c, c < C, &db, &dg);
+  if (threadIdx.x >= FIN_CH || c >= C) return;
   dbeta[c] = db;
   dgamma[c] = dg;
 }
@@ -274,24 +307,29 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_apply_kernel(
     const bf16x8* __restrict__ y, bf16x8* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ dbeta,
-    const float* __restrict__ dgamma, long n8, int c8, float invM) {
-  extern __shared__ float lds[];
-  float* s_a = lds;               // gamma*invstd
-  float* s_m = lds + c8 * 8;      // mean
-  float* s_is = lds + c8 * 16;    // invstd
-  float* s_db = lds + c8 * 24;    // dbeta/M
-  float* s_dg = lds + c8 * 32;    // dgamma/M
-  for (int i = threadIdx.x; i < c8 * 8; i += BN_BLOCK) {
-    s_a[i] = gamma[i] * invstd[i];
-    s_m[i] = mean[i];
-    s_is[i] = invstd[i];
-    s_db[i] = dbeta[i] * invM;
-    s_dg[i] = dgamma[i] * invM;
+    const float* __restrict__ dgamma, long M, int c8, float invM,
+    int rows_per_block) {
+  // Row-tile geometry: per-thread fixed channel octet, coefficients in
+  // registers (see bn_fwd_apply_kernel).
+  const int slots = c8;
+  const int rows_par = BN_BLOCK / slots;
+  const int slot = threadIdx.x % slots;
+  const int rsub = threadIdx.x / slots;
+  float a[8], m[8], is[8], db[8], dg[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    const int c = slot * 8 + k;
+    is[k] = invstd[c];
+    a[k] = gamma[c] * is[k];
+    m[k] = mean[c];
+    db[k] = dbeta[c] * invM;
+    dg[k] = dgamma[c] * invM;
   }
-  __syncthreads();
-  long stride = (long)gridDim.x * BN_BLOCK;
-  for (long i = (long)blockIdx.x * BN_BLOCK + threadIdx.x; i < n8; i += stride) {
-    int slot = (int)(i % c8) * 8;
+  long row0 = (long)blockIdx.x * rows_per_block;
+  long row_end = min(row0 + rows_per_block, M);
+  if (rsub >= rows_par) return;
+  for (long r = row0 + rsub; r < row_end; r += rows_par) {
+    long i = r * slots + slot;
     bf16x8 vdy = dy[i];
     bf16x8 vx = x[i];
     bf16x8 vy;
@@ -301,8 +339,8 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_apply_kernel(
     for (int k = 0; k < 8; ++k) {
       float g = b2f(vdy.h[k]);
       if (RELU && b2f(vy.h[k]) <= 0.f) g = 0.f;
-      float xh = (b2f(vx.h[k]) - s_m[slot + k]) * s_is[slot + k];
-      float d = s_a[slot + k] * (g - s_db[slot + k] - xh * s_dg[slot + k]);
+      float xh = (b2f(vx.h[k]) - m[k]) * is[k];
+      float d = a[k] * (g - db[k] - xh * dg[k]);
       o.h[k] = f2b(d);
     }
     dx[i] = o;
@@ -351,7 +389,8 @@ std::vector<at::Tensor> bn_fwd_train(
   hipLaunchKernelGGL(bn_fwd_stats_kernel, dim3(nblocks), dim3(BN_BLOCK), 0,
                      stream, (const bf16x8*)x.data_ptr(),
                      part.data_ptr<float>(), M, c8, rpb);
-  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + FIN_CH - 1) / FIN_CH),
+                     dim3(FIN_PPAR * FIN_CH),
                      0, stream, part.data_ptr<float>(), nblocks,
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -359,18 +398,15 @@ std::vector<at::Tensor> bn_fwd_train(
                      running_mean ? running_mean->data_ptr<float>() : nullptr,
                      running_var ? running_var->data_ptr<float>() : nullptr,
                      M, (int)C, (float)eps, (float)momentum);
-  long n8 = M * c8;
-  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
-  size_t lds_bytes = (size_t)C * 2 * sizeof(float);
   const bf16x8* resp =
       residual ? (const bf16x8*)residual->data_ptr() : nullptr;
   auto launch_apply = [&](auto relu_t, auto res_t) {
     hipLaunchKernelGGL((bn_fwd_apply_kernel<decltype(relu_t)::value,
                                             decltype(res_t)::value>),
-                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       dim3(nblocks), dim3(BN_BLOCK), 0, stream,
                        (const bf16x8*)x.data_ptr(), resp,
                        (bf16x8*)y.data_ptr(), scale.data_ptr<float>(),
-                       shift.data_ptr<float>(), n8, c8);
+                       shift.data_ptr<float>(), M, c8, rpb);
   };
   if (relu && residual) launch_apply(std::true_type{}, std::true_type{});
   else if (relu) launch_apply(std::true_type{}, std::false_type{});
@@ -391,18 +427,17 @@ at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> residual,
   auto scale = (gamma * invstd).contiguous();
   auto shift = (beta - running_mean * scale).contiguous();
   auto y = at::empty_like(x);
-  long n8 = M * c8;
-  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
-  size_t lds_bytes = (size_t)C * 2 * sizeof(float);
+  int rpb = pick_row_tiles(M, c8);
+  int nblocks = (int)((M + rpb - 1) / rpb);
   const bf16x8* resp =
       residual ? (const bf16x8*)residual->data_ptr() : nullptr;
   auto launch_apply = [&](auto relu_t, auto res_t) {
     hipLaunchKernelGGL((bn_fwd_apply_kernel<decltype(relu_t)::value,
                                             decltype(res_t)::value>),
-                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       dim3(nblocks), dim3(BN_BLOCK), 0, stream,
                        (const bf16x8*)x.data_ptr(), resp,
                        (bf16x8*)y.data_ptr(), scale.data_ptr<float>(),
-                       shift.data_ptr<float>(), n8, c8);
+                       shift.data_ptr<float>(), M, c8, rpb);
   };
   if (relu && residual) launch_apply(std::true_type{}, std::true_type{});
   else if (relu) launch_apply(std::true_type{}, std::false_type{});
@@ -445,22 +480,21 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   else if (relu) launch_reduce(std::true_type{}, std::false_type{});
   else if (needs_dres) launch_reduce(std::false_type{}, std::true_type{});
   else launch_reduce(std::false_type{}, std::false_type{});
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + FIN_CH - 1) / FIN_CH),
+                     dim3(FIN_PPAR * FIN_CH),
                      0, stream, part.data_ptr<float>(), nblocks,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
                      (int)C);
-  long n8 = M * c8;
-  int grid = (int)std::min<long>((n8 + BN_BLOCK - 1) / BN_BLOCK, 2048);
-  size_t lds_bytes = (size_t)C * 5 * sizeof(float);
   auto launch_apply = [&](auto relu_t) {
     hipLaunchKernelGGL((bn_bwd_apply_kernel<decltype(relu_t)::value>),
-                       dim3(grid), dim3(BN_BLOCK), lds_bytes, stream,
+                       dim3(nblocks), dim3(BN_BLOCK), 0, stream,
                        (const bf16x8*)x.data_ptr(),
                        (const bf16x8*)dy.data_ptr(),
                        (const bf16x8*)y.data_ptr(), (bf16x8*)dx.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                       dgamma.data_ptr<float>(), n8, c8, 1.f / (float)M);
+                       dgamma.data_ptr<float>(), M, c8, 1.f / (float)M,
+                       rpb);
   };
   if (relu) launch_apply(std::true_type{});
   else launch_apply(std::false_type{});
