@@ -88,6 +88,10 @@ def main():
     ap.add_argument("--config", required=True)
     ap.add_argument("--epochs", type=int, default=1)
     ap.add_argument("--samples", type=int, default=2048)
+    ap.add_argument("--max-steps", type=int, default=0,
+                    help="stop each epoch after N steps (0 = full epoch)")
+    ap.add_argument("--batch", type=int, default=0,
+                    help="override the config batch size")
     args = ap.parse_args()
     with open(args.config) as f:
         cfg = yaml.safe_load(f)
@@ -104,7 +108,7 @@ def main():
             optimizer_kwargs=cfg.get("optimizer", {"lr": 1e-3}),
         ),
         loss=nn.CrossEntropyLoss(),
-        batch_size_per_device=run.get("batch_size", 32),
+        batch_size_per_device=args.batch or run.get("batch_size", 32),
         grad_accum_steps=run.get("grad_accum", 1),
         grad_clip=grad_clip,
         gpu=run.get("gpu", False),
@@ -128,16 +132,18 @@ def main():
     for epoch in range(args.epochs):
         if sampler is not None:
             sampler.set_epoch(epoch)
-        train(stoke_obj, loader)
+        train(stoke_obj, loader, args.max_steps)
         stoke_obj.print_ema_loss()
     # predict loop
-    correct = predict(stoke_obj, loader)
+    correct = predict(stoke_obj, loader, args.max_steps)
     stoke_obj.print(f"train accuracy: {correct:.3f}")
 
 
-def train(stoke_obj, loader):
+def train(stoke_obj, loader, max_steps=0):
     stoke_obj.model_access.train()
-    for x, y in loader:
+    for i, (x, y) in enumerate(loader):
+        if max_steps and i >= max_steps:
+            break
         out = stoke_obj.model(x)
         loss = stoke_obj.loss(out, y)
         stoke_obj.print_mean_accumulated_synced_loss()
@@ -145,11 +151,13 @@ def train(stoke_obj, loader):
         stoke_obj.step()
 
 
-def predict(stoke_obj, loader):
+def predict(stoke_obj, loader, max_steps=0):
     stoke_obj.model_access.eval()
     hits = n = 0
     with torch.no_grad():
-        for x, y in loader:
+        for i, (x, y) in enumerate(loader):
+            if max_steps and i >= max_steps:
+                break
             out = stoke_obj.model(x)
             hits += (out.argmax(-1) == y).sum().item()
             n += len(y)
