@@ -183,17 +183,8 @@ def gpt2_medium(vocab=50257, max_seq=1024):
 # ---------------------------------------------------------------------------
 # Llama-3-shape decoder (RMSNorm, RoPE, SwiGLU, GQA) — sized for 8B default
 # ---------------------------------------------------------------------------
-class RMSNorm(nn.Module):
-    def __init__(self, d, eps=1e-5):
-        super().__init__()
-        self.weight = nn.Parameter(torch.ones(d))
-        self.eps = eps
-
-    def forward(self, x):
-        dt = x.dtype
-        x = x.float()
-        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
-        return (x * self.weight.float()).to(dt)
+# HIP-fused on bf16 GPU inputs, eager fp32 composition elsewhere
+from stoke.nn import StokeRMSNorm as RMSNorm  # noqa: E402
 
 
 def _rope_cache(seq, hd, device, base=500000.0):

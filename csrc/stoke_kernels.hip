@@ -414,12 +414,20 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                                at::Tensor mean, at::Tensor invstd,
                                at::Tensor gamma, bool relu, bool needs_dres);
 
+// Fused RMSNorm kernels (csrc/fused_rmsnorm.hip)
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
+                                    at::Tensor invr);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN forward (eval)");
   m.def("bn_bwd", &bn_bwd,
         "fused NHWC bf16 BN backward: reduce(+relu mask) + dx(+dres)");
+  m.def("rmsnorm_fwd", &rmsnorm_fwd,
+        "fused bf16 RMSNorm forward: y = x*rsqrt(mean(x^2)+eps)*w");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

@@ -31,7 +31,8 @@ setup(
     ext_modules=[
         cpp_extension.CUDAExtension(
             name="stoke._C",
-            sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip"],
+            sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip",
+                     "csrc/fused_rmsnorm.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],
