@@ -194,15 +194,8 @@ def _rope_cache(seq, hd, device, base=500000.0):
     return freqs.cos(), freqs.sin()
 
 
-def _apply_rope(x, cos, sin):
-    # x: [B, H, S, D]
-    x1, x2 = x[..., 0::2], x[..., 1::2]
-    c = cos[None, None, : x.shape[2]]
-    s = sin[None, None, : x.shape[2]]
-    out = torch.empty_like(x)
-    out[..., 0::2] = x1 * c - x2 * s
-    out[..., 1::2] = x2 * c + x1 * s
-    return out
+# HIP-fused on bf16 GPU inputs ([B, S, H, Dh] layout), eager elsewhere
+from stoke.nn.rope import apply_rope  # noqa: E402
 
 
 class LlamaBlock(nn.Module):
@@ -223,11 +216,13 @@ class LlamaBlock(nn.Module):
     def forward(self, x, cos, sin):
         B, S, D = x.shape
         h = self.attn_norm(x)
-        q = self.wq(h).view(B, S, self.nh, self.hd).transpose(1, 2)
-        k = self.wk(h).view(B, S, self.nkv, self.hd).transpose(1, 2)
+        # RoPE on the contiguous [B, S, H, Dh] projections (one fused kernel
+        # each), then the attention transpose
+        q = apply_rope(self.wq(h).view(B, S, self.nh, self.hd), cos, sin)
+        k = apply_rope(self.wk(h).view(B, S, self.nkv, self.hd), cos, sin)
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
         v = self.wv(h).view(B, S, self.nkv, self.hd).transpose(1, 2)
-        q = _apply_rope(q, cos, sin)
-        k = _apply_rope(k, cos, sin)
         a = F.scaled_dot_product_attention(
             q, k, v, is_causal=True, enable_gqa=(self.nkv != self.nh)
         )

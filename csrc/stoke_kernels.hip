@@ -419,6 +419,10 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
                                     at::Tensor invr);
 
+// Fused RoPE kernel (csrc/fused_rope.hip)
+at::Tensor rope_apply(at::Tensor x, at::Tensor cosc, at::Tensor sinc,
+                      bool conj);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
@@ -428,6 +432,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd,
         "fused bf16 RMSNorm forward: y = x*rsqrt(mean(x^2)+eps)*w");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
+  m.def("rope_apply", &rope_apply,
+        "fused bf16 rotary embedding (conj=true for the backward rotation)");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

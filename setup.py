@@ -32,7 +32,8 @@ setup(
         cpp_extension.CUDAExtension(
             name="stoke._C",
             sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip",
-                     "csrc/fused_rmsnorm.hip"],
+                     "csrc/fused_rmsnorm.hip",
+                     "csrc/fused_rope.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],
