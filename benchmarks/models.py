@@ -229,7 +229,9 @@ class LlamaBlock(nn.Module):
         a = a.transpose(1, 2).reshape(B, S, -1)
         x = x + self.wo(a)
         h = self.mlp_norm(x)
-        return x + self.w_down(F.silu(self.w_gate(h)) * self.w_up(h))
+        from stoke.nn import swiglu
+
+        return x + self.w_down(swiglu(self.w_gate(h), self.w_up(h)))
 
 
 class Llama(nn.Module):

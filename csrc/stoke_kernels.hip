@@ -271,8 +271,24 @@ struct AdamWFunctor {
 
 // AdamW with bf16 model params + fp32 master params/state, bf16 grads
 // (the FSDP bf16 flat-shard path: depth 5 = [bf16 p, bf16 g, fp32 m, fp32 v,
-// fp32 master]).
+// fp32 master]).  Vectorized 8-wide: bf16x8 (16 B) for p/g, 2x float4 for
+// m/v/w — the scalar version measured 19% of a Llama-8B step (2-byte loads).
+union adam_bf16x8 {
+  uint4 u4;
+  unsigned short h[8];
+};
+
 struct AdamWBF16Functor {
+  __device__ __forceinline__ float upd(float gg, float& mm, float& vn,
+                                       float pw, float lr, float beta1,
+                                       float beta2, float eps, float wd,
+                                       float bc1, float bc2) const {
+    mm = beta1 * mm + (1.f - beta1) * gg;
+    vn = beta2 * vn + (1.f - beta2) * gg * gg;
+    float denom = sqrtf(vn / bc2) + eps;
+    return pw - lr * ((mm / bc1) / denom + wd * pw);
+  }
+
   template <typename Meta>
   __device__ void operator()(const Meta& meta, int t, int start, int len,
                              float lr, float beta1, float beta2, float eps,
@@ -285,14 +301,63 @@ struct AdamWBF16Functor {
     float* v = ((float*)meta.addr[3][t]) + start;
     float* w = ((float*)meta.addr[4][t]) + start;  // fp32 master copy
     const float inv = inv_scale ? *inv_scale : 1.f;
-    for (int i = threadIdx.x; i < len; i += BLOCK) {
+    const bool aligned = ((reinterpret_cast<uintptr_t>(p) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(g) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(m) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(v) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(w) & 15) == 0);
+    const int n8 = aligned ? (len & ~7) : 0;
+    for (int i = threadIdx.x * 8; i < n8; i += BLOCK * 8) {
+      adam_bf16x8 gv;
+      gv.u4 = *(const uint4*)(g + i);
+      float4v m0 = *(const float4v*)(m + i);
+      float4v m1 = *(const float4v*)(m + i + 4);
+      float4v v0 = *(const float4v*)(v + i);
+      float4v v1 = *(const float4v*)(v + i + 4);
+      float4v w0 = *(const float4v*)(w + i);
+      float4v w1 = *(const float4v*)(w + i + 4);
+      adam_bf16x8 pv;
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        __hip_bfloat16_raw r;
+        r.x = gv.h[k];
+        float gg = __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&r)) * inv;
+        float mm = m0[k], vn = v0[k];
+        float pw = upd(gg, mm, vn, w0[k], lr, beta1, beta2, eps,
+                       weight_decay, bc1, bc2);
+        m0[k] = mm;
+        v0[k] = vn;
+        w0[k] = pw;
+        __hip_bfloat16 hb = __float2bfloat16(pw);
+        pv.h[k] = *reinterpret_cast<unsigned short*>(&hb);
+      }
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        __hip_bfloat16_raw r;
+        r.x = gv.h[4 + k];
+        float gg = __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&r)) * inv;
+        float mm = m1[k], vn = v1[k];
+        float pw = upd(gg, mm, vn, w1[k], lr, beta1, beta2, eps,
+                       weight_decay, bc1, bc2);
+        m1[k] = mm;
+        v1[k] = vn;
+        w1[k] = pw;
+        __hip_bfloat16 hb = __float2bfloat16(pw);
+        pv.h[4 + k] = *reinterpret_cast<unsigned short*>(&hb);
+      }
+      *(uint4*)(p + i) = pv.u4;
+      *(float4v*)(m + i) = m0;
+      *(float4v*)(m + i + 4) = m1;
+      *(float4v*)(v + i) = v0;
+      *(float4v*)(v + i + 4) = v1;
+      *(float4v*)(w + i) = w0;
+      *(float4v*)(w + i + 4) = w1;
+    }
+    for (int i = n8 + threadIdx.x; i < len; i += BLOCK) {
       float gg = __bfloat162float(g[i]) * inv;
-      float mm = beta1 * m[i] + (1.f - beta1) * gg;
-      float vn = beta2 * v[i] + (1.f - beta2) * gg * gg;
-      float denom = sqrtf(vn / bc2) + eps;
-      float pw = w[i];
-      float update = (mm / bc1) / denom + weight_decay * pw;
-      pw = pw - lr * update;
+      float mm = m[i], vn = v[i];
+      float pw = upd(gg, mm, vn, w[i], lr, beta1, beta2, eps, weight_decay,
+                     bc1, bc2);
       w[i] = pw;
       p[i] = __float2bfloat16(pw);
       m[i] = mm;
@@ -423,6 +488,10 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
 at::Tensor rope_apply(at::Tensor x, at::Tensor cosc, at::Tensor sinc,
                       bool conj);
 
+// Fused SwiGLU kernels (csrc/fused_swiglu.hip)
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
@@ -434,6 +503,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("rope_apply", &rope_apply,
         "fused bf16 rotary embedding (conj=true for the backward rotation)");
+  m.def("swiglu_fwd", &swiglu_fwd, "fused bf16 silu(g)*u");
+  m.def("swiglu_bwd", &swiglu_bwd, "fused bf16 SwiGLU backward");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

@@ -33,7 +33,8 @@ setup(
             name="stoke._C",
             sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip",
                      "csrc/fused_rmsnorm.hip",
-                     "csrc/fused_rope.hip"],
+                     "csrc/fused_rope.hip",
+                     "csrc/fused_swiglu.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

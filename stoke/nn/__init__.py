@@ -1,3 +1,4 @@
 from stoke.nn.act_ckpt import apply_activation_checkpointing  # noqa: F401
 from stoke.nn.fused_bn import FusedBNAct2d  # noqa: F401
 from stoke.nn.rmsnorm import StokeRMSNorm  # noqa: F401
+from stoke.nn.swiglu import swiglu  # noqa: F401

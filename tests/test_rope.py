@@ -60,3 +60,40 @@ def test_gpu_rope_backward_is_inverse_rotation():
     _eager_rope(x32, cos, sin).backward(gy)
     err = (x16.grad.float() - x32.grad).abs().max().item()
     assert err / (x32.grad.abs().max().item() + 1e-6) < 0.03, err
+
+
+# ------------------------------------------------------------------ SwiGLU
+def test_cpu_swiglu_matches_eager():
+    import torch.nn.functional as F
+
+    from stoke.nn import swiglu
+
+    torch.manual_seed(0)
+    g = torch.randn(64, 32, requires_grad=True)
+    u = torch.randn(64, 32, requires_grad=True)
+    y = swiglu(g, u)
+    assert torch.allclose(y, F.silu(g) * u, atol=1e-6)
+    y.pow(2).sum().backward()
+    assert g.grad is not None and u.grad is not None
+
+
+@pytest.mark.gpu
+def test_gpu_swiglu_vs_fp32():
+    import torch.nn.functional as F
+
+    from stoke.nn import swiglu
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.manual_seed(3)
+    g16 = torch.randn(1000, 256, device="cuda").bfloat16().requires_grad_(True)
+    u16 = torch.randn(1000, 256, device="cuda").bfloat16().requires_grad_(True)
+    gy = torch.randn(1000, 256, device="cuda")
+    swiglu(g16, u16).backward(gy.bfloat16())
+
+    g32 = g16.detach().float().requires_grad_(True)
+    u32 = u16.detach().float().requires_grad_(True)
+    (F.silu(g32) * u32).backward(gy)
+    for got, want in ((g16.grad, g32.grad), (u16.grad, u32.grad)):
+        err = (got.float() - want).abs().max().item()
+        assert err / (want.abs().max().item() + 1e-6) < 0.05, err
