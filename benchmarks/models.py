@@ -135,19 +135,32 @@ def resnet152(num_classes=1000, small_input=False):
 # GPT-2-shape decoder (learned positions, LayerNorm, GELU MLP)
 # ---------------------------------------------------------------------------
 class GPT2Block(nn.Module):
+    """Pre-LN GPT-2 block with fused QKV + causal flash SDPA.
+
+    (An earlier version used nn.MultiheadAttention with an additive mask,
+    which falls off the SDPA fast path and materializes S x S weights.)
+    """
+
     def __init__(self, d, nh, dropout=0.0):
         super().__init__()
+        self.nh = nh
         self.ln1 = nn.LayerNorm(d)
-        self.attn = nn.MultiheadAttention(d, nh, dropout=dropout, batch_first=True)
+        self.qkv = nn.Linear(d, 3 * d)
+        self.proj = nn.Linear(d, d)
         self.ln2 = nn.LayerNorm(d)
         self.mlp = nn.Sequential(
             nn.Linear(d, 4 * d), nn.GELU(), nn.Linear(4 * d, d)
         )
 
-    def forward(self, x, attn_mask=None):
+    def forward(self, x):
+        B, S, D = x.shape
         h = self.ln1(x)
-        a, _ = self.attn(h, h, h, attn_mask=attn_mask, need_weights=False)
-        x = x + a
+        q, k, v = self.qkv(h).split(D, dim=-1)
+        q = q.view(B, S, self.nh, -1).transpose(1, 2)
+        k = k.view(B, S, self.nh, -1).transpose(1, 2)
+        v = v.view(B, S, self.nh, -1).transpose(1, 2)
+        a = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        x = x + self.proj(a.transpose(1, 2).reshape(B, S, D))
         return x + self.mlp(self.ln2(x))
 
 
@@ -168,11 +181,8 @@ class GPT2(nn.Module):
         B, S = idx.shape
         pos = torch.arange(S, device=idx.device)
         x = self.wte(idx) + self.wpe(pos)[None]
-        mask = torch.triu(
-            torch.full((S, S), float("-inf"), device=idx.device), diagonal=1
-        )
         for blk in self.blocks:
-            x = blk(x, attn_mask=mask)
+            x = blk(x)
         return self.head(self.ln_f(x))
 
 
