@@ -159,7 +159,7 @@ class DeepspeedActivationCheckpointingConfig:
 
 @attr.s(auto_attribs=True)
 class DeepspeedFlopsConfig:
-    """Flops profiler knobs (reference ``configs.py:251-279``); see stoke.utils.flops."""
+    """Flops profiler knobs (reference ``configs.py:251-279``); see stoke.utils.FlopsProfiler."""
 
     detailed: bool = True
     module_depth: int = -1

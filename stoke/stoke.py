@@ -130,6 +130,19 @@ class Stoke:
         self._rolling_mean_loss = self._set_loss_to_zero()
         self._rolling_loss_steps = 0
         self._status.set_post_init_values(world_size=self.world_size)
+        # Flops profiler (in-house DeepspeedFlopsConfig equivalent): hooks
+        # count forward FLOPs until profile_step optimizer steps, then print
+        self._flops_profiler = None
+        self._flops_cfg = None
+        if (
+            self._status.is_distributed_deepspeed
+            and self._status.deepspeed_config.flops_profiler is not None
+        ):
+            from stoke.utils import FlopsProfiler
+
+            self._flops_cfg = self._status.deepspeed_config.flops_profiler
+            self._flops_profiler = FlopsProfiler(self.model_access)
+            self._flops_profiler.start_profile()
         if self._verbose:
             self.print(msg=self._status)
 
@@ -419,6 +432,17 @@ class Stoke:
                 )
             self._reset()
             self._optimizer_steps += 1
+            if (
+                self._flops_profiler is not None
+                and self._optimizer_steps >= self._flops_cfg.profile_step
+            ):
+                self._flops_profiler.print_model_profile(
+                    top_modules=self._flops_cfg.top_modules,
+                    detailed=self._flops_cfg.detailed,
+                    output_file=self._flops_cfg.output_file,
+                )
+                self._flops_profiler.stop_profile()
+                self._flops_profiler = None
         elif self.is_deepspeed:
             step_cm = (
                 self._runner.step_context(self._optimizer)

@@ -100,3 +100,88 @@ def make_folder(path: str):
     """Create a directory (and parents) if it does not already exist."""
     if not os.path.isdir(path):
         os.makedirs(path, exist_ok=True)
+
+
+class FlopsProfiler:
+    """Per-module FLOPs / parameter profiler (the in-house equivalent of the
+    DeepSpeed flops profiler the reference exposed via ``DeepspeedFlopsConfig``,
+    reference ``configs.py:251-279`` -> ``distributed.py:985-1004``).
+
+    Counts multiply-accumulates as 2 FLOPs via forward hooks on leaf modules
+    (Linear / Conv2d / BatchNorm-like / activations); functional ops that do
+    not pass through a module (e.g. ``F.scaled_dot_product_attention``) are
+    not seen — the printed total is a floor, like most hook-based profilers.
+    """
+
+    def __init__(self, model: torch.nn.Module):
+        self.model = model
+        self._handles: List = []
+        self.flops: Dict[str, float] = {}
+        self.params: Dict[str, int] = {}
+
+    @staticmethod
+    def _module_flops(mod: torch.nn.Module, inp, out) -> float:
+        if isinstance(mod, torch.nn.Linear):
+            return 2.0 * out.numel() * mod.in_features
+        if isinstance(mod, torch.nn.Conv2d):
+            k = mod.kernel_size[0] * mod.kernel_size[1]
+            cin = mod.in_channels // mod.groups
+            return 2.0 * out.numel() * cin * k
+        if isinstance(mod, (torch.nn.BatchNorm2d, torch.nn.LayerNorm)):
+            return 4.0 * out.numel()
+        if mod.__class__.__name__ in ("FusedBNAct2d", "StokeRMSNorm"):
+            return 4.0 * out.numel()
+        if isinstance(mod, (torch.nn.ReLU, torch.nn.GELU, torch.nn.SiLU)):
+            return float(out.numel())
+        if isinstance(mod, torch.nn.Embedding):
+            return 0.0
+        return 0.0
+
+    def start_profile(self):
+        self.flops.clear()
+        self.params.clear()
+
+        def make_hook(name):
+            def hook(mod, inp, out):
+                o = out[0] if isinstance(out, (list, tuple)) else out
+                if isinstance(o, torch.Tensor):
+                    self.flops[name] = self.flops.get(name, 0.0) + \
+                        self._module_flops(mod, inp, o)
+            return hook
+
+        for name, mod in self.model.named_modules():
+            if len(list(mod.children())) == 0:  # leaves only
+                self.params[name] = sum(p.numel() for p in mod.parameters())
+                self._handles.append(mod.register_forward_hook(make_hook(name)))
+
+    def stop_profile(self):
+        for h in self._handles:
+            h.remove()
+        self._handles.clear()
+
+    def get_total_flops(self) -> float:
+        return sum(self.flops.values())
+
+    def get_total_params(self) -> int:
+        return sum(self.params.values())
+
+    def print_model_profile(self, top_modules: int = 1, detailed: bool = True,
+                            output_file: Optional[str] = None):
+        lines = [
+            f"FLOPs profile: total fwd flops {self.get_total_flops()/1e9:.2f} GFLOPs, "
+            f"params {self.get_total_params()/1e6:.2f} M"
+        ]
+        if detailed:
+            ranked = sorted(self.flops.items(), key=lambda kv: -kv[1])
+            for name, fl in ranked[: max(top_modules, 1) * 10]:
+                lines.append(
+                    f"  {name:60s} {fl/1e9:10.3f} GFLOPs "
+                    f"{self.params.get(name, 0)/1e6:8.2f} M params"
+                )
+        text = "\n".join(lines)
+        if output_file:
+            with open(output_file, "w") as f:
+                f.write(text + "\n")
+        else:
+            unrolled_print(lines)
+        return text

@@ -244,3 +244,22 @@ def test_properties_surface():
     s.print_num_model_parameters()
     s.dump_model_parameter_info()
     s.print_ema_loss()
+
+
+def test_flops_profiler_counts():
+    import torch as _t
+
+    from benchmarks.models import resnet18
+    from stoke.utils import FlopsProfiler
+
+    m = resnet18(num_classes=10, small_input=True)
+    fp = FlopsProfiler(m)
+    fp.start_profile()
+    m(_t.randn(2, 3, 32, 32))
+    fp.stop_profile()
+    # ResNet-18 CIFAR-shape is ~0.56 GMACs/sample forward
+    per_sample = fp.get_total_flops() / 2
+    assert 0.8e9 < per_sample < 1.5e9
+    assert abs(fp.get_total_params() - sum(p.numel() for p in m.parameters())) < 1e4
+    text = fp.print_model_profile(detailed=True)
+    assert "GFLOPs" in text
