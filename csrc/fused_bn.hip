@@ -67,6 +67,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_stats_kernel(
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float acc2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   if (rsub < rows_par) {
+    #pragma unroll 2
     for (long r = row0 + rsub; r < row_end; r += rows_par) {
       bf16x8 v = x[r * slots + slot];
       #pragma unroll
@@ -195,6 +196,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_apply_kernel(
   long row0 = (long)blockIdx.x * rows_per_block;
   long row_end = min(row0 + rows_per_block, M);
   if (rsub >= rows_par) return;
+  #pragma unroll 2
   for (long r = row0 + rsub; r < row_end; r += rows_par) {
     long i = r * slots + slot;
     bf16x8 v = x[i];
@@ -238,6 +240,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_reduce_kernel(
   float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   if (rsub < rows_par) {
+    #pragma unroll 2
     for (long r = row0 + rsub; r < row_end; r += rows_par) {
       long i = r * slots + slot;
       bf16x8 vdy = dy[i];
@@ -328,6 +331,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_apply_kernel(
   long row0 = (long)blockIdx.x * rows_per_block;
   long row_end = min(row0 + rows_per_block, M);
   if (rsub >= rows_par) return;
+  #pragma unroll 2
   for (long r = row0 + rsub; r < row_end; r += rows_par) {
     long i = r * slots + slot;
     bf16x8 vdy = dy[i];
