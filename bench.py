@@ -36,6 +36,8 @@ def parse_args():
     p.add_argument("--seq", type=int, default=0, help="seq len for LM benches")
     p.add_argument("--cpu", action="store_true", help="tiny CPU plumbing run")
     p.add_argument("--bucket-mb", type=int, default=64)
+    p.add_argument("--fp8", action="store_true",
+                   help="fp8 (e4m3/e5m2) GEMMs for the LM benches (CDNA4)")
     return p.parse_args()
 
 
@@ -105,6 +107,8 @@ def main():
         dtype = "bf16"
         mname = "llama3-8b"
 
+    if args.fp8:
+        dtype = "fp8"  # stretch datapoint only — never the headline metric
     if args.model in ("gpt2-oss", "llama-fsdp"):
         def loss_fn(logits, target):
             return torch.nn.functional.cross_entropy(
@@ -117,6 +121,12 @@ def main():
         # fp32 module.
         if not (args.model == "llama-fsdp" and distributed):
             model = model.bfloat16()
+            if args.fp8:
+                from stoke.nn import convert_linears_to_fp8
+
+                nfp8 = convert_linears_to_fp8(model)
+                if rank == 0:
+                    print(f"# fp8: converted {nfp8} Linear layers")
     else:
         loss_fn = torch.nn.CrossEntropyLoss()
 

@@ -97,3 +97,27 @@ def test_gpu_swiglu_vs_fp32():
     for got, want in ((g16.grad, g32.grad), (u16.grad, u32.grad)):
         err = (got.float() - want).abs().max().item()
         assert err / (want.abs().max().item() + 1e-6) < 0.05, err
+
+
+# --------------------------------------------------------------------- FP8
+@pytest.mark.gpu
+def test_gpu_fp8_linear_if_available():
+    from stoke.nn import FP8Linear, fp8_available
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    if not fp8_available():
+        pytest.skip("fp8 _scaled_mm unavailable on this build")
+    torch.manual_seed(4)
+    lin = FP8Linear(256, 512, bias=True).cuda().bfloat16()
+    x = torch.randn(64, 256, device="cuda").bfloat16().requires_grad_(True)
+    y = lin(x)
+    assert y.shape == (64, 512) and y.dtype == torch.bfloat16
+    # fp8 per-tensor scaling keeps relative error within a few percent
+    ref = torch.nn.functional.linear(x.float(), lin.weight.float(),
+                                     lin.bias.float())
+    rel = (y.float() - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.08, f"fp8 fwd rel err {rel}"
+    y.float().pow(2).mean().backward()
+    assert x.grad is not None and lin.weight.grad is not None
+    assert torch.isfinite(x.grad).all()
