@@ -143,11 +143,13 @@ class GPT2Block(nn.Module):
 
     def __init__(self, d, nh, dropout=0.0):
         super().__init__()
+        from stoke.nn import StokeLayerNorm
+
         self.nh = nh
-        self.ln1 = nn.LayerNorm(d)
+        self.ln1 = StokeLayerNorm(d)
         self.qkv = nn.Linear(d, 3 * d)
         self.proj = nn.Linear(d, d)
-        self.ln2 = nn.LayerNorm(d)
+        self.ln2 = StokeLayerNorm(d)
         self.mlp = nn.Sequential(
             nn.Linear(d, 4 * d), nn.GELU(), nn.Linear(4 * d, d)
         )
@@ -171,8 +173,10 @@ class GPT2(nn.Module):
         super().__init__()
         self.wte = nn.Embedding(vocab, d)
         self.wpe = nn.Embedding(max_seq, d)
+        from stoke.nn import StokeLayerNorm
+
         self.blocks = nn.ModuleList([GPT2Block(d, nh) for _ in range(nlayer)])
-        self.ln_f = nn.LayerNorm(d)
+        self.ln_f = StokeLayerNorm(d)
         self.head = nn.Linear(d, vocab, bias=False)
         self.head.weight = self.wte.weight  # tied
         self.max_seq = max_seq

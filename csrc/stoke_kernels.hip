@@ -492,6 +492,13 @@ at::Tensor rope_apply(at::Tensor x, at::Tensor cosc, at::Tensor sinc,
 at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u);
 
+// Fused LayerNorm kernels (csrc/fused_layernorm.hip)
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                                      double eps);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor x, at::Tensor dy,
+                                      at::Tensor w, at::Tensor mean,
+                                      at::Tensor invstd);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
@@ -505,6 +512,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused bf16 rotary embedding (conj=true for the backward rotation)");
   m.def("swiglu_fwd", &swiglu_fwd, "fused bf16 silu(g)*u");
   m.def("swiglu_bwd", &swiglu_bwd, "fused bf16 SwiGLU backward");
+  m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

@@ -34,7 +34,8 @@ setup(
             sources=["csrc/stoke_kernels.hip", "csrc/fused_bn.hip",
                      "csrc/fused_rmsnorm.hip",
                      "csrc/fused_rope.hip",
-                     "csrc/fused_swiglu.hip"],
+                     "csrc/fused_swiglu.hip",
+                     "csrc/fused_layernorm.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],
