@@ -35,3 +35,13 @@ def test_example_gpu_bf16_config_runs():
         pytest.skip("no GPU")
     r = _run("gpu_bf16.yaml")
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_llm_example_cpu_runs():
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "llm", "train.py"),
+         "--cpu", "--layers", "2", "--dim", "256", "--steps", "3"],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "done: 3 steps" in r.stdout
