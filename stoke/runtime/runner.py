@@ -187,10 +187,15 @@ class StokeRunner:
         if convert_bn:
             model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
         # In-house activation checkpointing (the reference delegated this to
-        # the DeepSpeed engine via DeepspeedActivationCheckpointingConfig)
-        if (
-            s.is_distributed_deepspeed
-            and s.deepspeed_config.activation_checkpointing is not None
+        # the DeepSpeed engine via DeepspeedActivationCheckpointingConfig).
+        # The config object defaults to an INSTANCE (reference parity), so
+        # presence alone is not the switch: recompute is enabled by the
+        # explicit knobs (partition_activations / number_checkpoints), which
+        # is when the reference's DeepSpeed engine would actually trade
+        # memory for recompute.
+        if s.is_distributed_deepspeed and (
+            (ac := s.deepspeed_config.activation_checkpointing) is not None
+            and (ac.partition_activations or ac.number_checkpoints)
         ):
             from stoke.nn import apply_activation_checkpointing
 
