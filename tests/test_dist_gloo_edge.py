@@ -147,3 +147,50 @@ def test_oss_with_offloaded_state():
     torch.multiprocessing.spawn(
         _oss_offload_worker, args=(2, free_port(), 3), nprocs=2, join=True
     )
+
+
+def _unused_param_worker(rank, world, port):
+    pg = init_gloo(rank, world, port)
+    from stoke.ddp import StokeDDPModule
+
+    class Branchy(nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(0)
+            self.used = nn.Linear(8, 4)
+            self.never = nn.Linear(8, 4)  # produces no grad
+
+        def forward(self, x):
+            return self.used(x)
+
+    model = Branchy()
+    ddp = StokeDDPModule(model, pg=pg, bucket_cap_mb=1,
+                         find_unused_parameters=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    for step in range(2):
+        x, y = _data(rank * 7 + step)
+        nn.CrossEntropyLoss()(ddp(x), y).backward()
+        ddp.finish_backward()  # must not hang on `never`'s missing grads
+        opt.step()
+        opt.zero_grad()
+    # used params stay synced; never-params unchanged and identical
+    for p in model.parameters():
+        flat = p.detach().reshape(-1).clone()
+        ref = flat.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(flat, ref, atol=1e-6)
+    dist.destroy_process_group()
+
+
+def test_ddp_unused_parameters_no_hang():
+    torch.multiprocessing.spawn(
+        _unused_param_worker, args=(2, free_port()), nprocs=2, join=True
+    )
+
+
+def test_ddp_accum_with_bucket_view():
+    from tests.test_dist_gloo import _ddp_worker
+
+    torch.multiprocessing.spawn(
+        _ddp_worker, args=(2, free_port(), 2, 3, True), nprocs=2, join=True
+    )
