@@ -217,10 +217,16 @@ class OSSOptimizer(torch.optim.Optimizer):
         full_state = {}
         for shard in gathered:
             full_state.update(shard)
+        # Emit FULL hyperparameters per group (inner-optimizer defaults +
+        # per-group overrides): the outer param_groups only carry the kwargs
+        # the user passed, and a checkpoint that omitted e.g. Adam's betas
+        # could not be loaded into a plain torch optimizer at another world
+        # size (the world-size-independence contract).
         return {
             "state": full_state,
             "param_groups": [
                 {
+                    **self.optim.defaults,
                     **{k: v for k, v in g.items() if k != "params"},
                     "params": [index_of[id(p)] for p in g["params"]],
                 }

@@ -1,0 +1,182 @@
+# -*- coding: utf-8 -*-
+"""More distributed hardening on gloo: FSDP gradient accumulation, OSS with
+multiple param groups, world-size-independent checkpoint reload."""
+
+import os
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from tests.test_dist_gloo import _data, _model, free_port, init_gloo
+
+
+def _fsdp_accum_worker(rank, world, port, steps, accum):
+    pg = init_gloo(rank, world, port)
+    from contextlib import nullcontext
+
+    from stoke.shard import StokeFSDPModule
+
+    model = _model()
+    fsdp = StokeFSDPModule(model, pg=pg, compute_dtype=torch.float32,
+                           reshard_after_forward=True, min_wrap_params=100)
+    opt = torch.optim.Adam(fsdp.parameters(), lr=0.01)
+    for step in range(steps):
+        for micro in range(accum):
+            x, y = _data(rank * 1000 + step * 13 + micro)
+            cm = fsdp.no_sync() if micro < accum - 1 else nullcontext()
+            with cm:
+                (nn.CrossEntropyLoss()(fsdp(x), y) / accum).backward()
+        fsdp.finish_backward()
+        opt.step()
+        opt.zero_grad()
+    ref_model = _model()
+    ref_opt = torch.optim.Adam(ref_model.parameters(), lr=0.01)
+    for step in range(steps):
+        ref_opt.zero_grad()
+        losses = [
+            nn.CrossEntropyLoss()(
+                ref_model(_data(r * 1000 + step * 13 + m)[0]),
+                _data(r * 1000 + step * 13 + m)[1],
+            )
+            for r in range(world)
+            for m in range(accum)
+        ]
+        (sum(losses) / (world * accum)).backward()
+        ref_opt.step()
+    sd = fsdp.full_state_dict()
+    for name, rp in ref_model.named_parameters():
+        assert torch.allclose(sd[name], rp.detach(), atol=2e-5), \
+            f"rank {rank}: FSDP accum {name} off {(sd[name]-rp.detach()).abs().max()}"
+    dist.destroy_process_group()
+
+
+def test_fsdp_grad_accum_no_sync():
+    torch.multiprocessing.spawn(
+        _fsdp_accum_worker, args=(2, free_port(), 2, 2), nprocs=2, join=True
+    )
+
+
+def _oss_groups_worker(rank, world, port):
+    pg = init_gloo(rank, world, port)
+    from stoke.ddp import StokeDDPModule
+    from stoke.shard import OSSOptimizer
+
+    model = _model()
+    ddp = StokeDDPModule(model, pg=pg)
+    params = list(model.parameters())
+    groups = [
+        {"params": params[: len(params) // 2], "lr": 0.05},
+        {"params": params[len(params) // 2 :], "lr": 0.005},
+    ]
+    opt = OSSOptimizer(groups, optim=torch.optim.SGD, pg=pg, lr=0.01)
+    for step in range(3):
+        x, y = _data(rank * 1000 + step)
+        nn.CrossEntropyLoss()(ddp(x), y).backward()
+        ddp.finish_backward()
+        opt.step()
+        opt.zero_grad()
+    ref_model = _model()
+    rparams = list(ref_model.parameters())
+    ref_opt = torch.optim.SGD(
+        [
+            {"params": rparams[: len(rparams) // 2], "lr": 0.05},
+            {"params": rparams[len(rparams) // 2 :], "lr": 0.005},
+        ],
+        lr=0.01,
+    )
+    for step in range(3):
+        ref_opt.zero_grad()
+        losses = [
+            nn.CrossEntropyLoss()(ref_model(_data(r * 1000 + step)[0]),
+                                  _data(r * 1000 + step)[1])
+            for r in range(world)
+        ]
+        (sum(losses) / world).backward()
+        ref_opt.step()
+    for p, r in zip(model.parameters(), ref_model.parameters()):
+        assert torch.allclose(p.detach(), r.detach(), atol=1e-5), \
+            f"rank {rank}: per-group lr mismatch {(p.detach()-r.detach()).abs().max()}"
+    dist.destroy_process_group()
+
+
+def test_oss_multiple_param_groups():
+    torch.multiprocessing.spawn(
+        _oss_groups_worker, args=(2, free_port()), nprocs=2, join=True
+    )
+
+
+def _save_w2_worker(rank, world, port, tmpdir):
+    pg = init_gloo(rank, world, port)
+    from stoke import io_ops
+    from stoke.ddp import StokeDDPModule
+    from stoke.shard import OSSOptimizer
+
+    class _R:
+        rank = pg.rank
+
+        def barrier(self):
+            pg.barrier()
+
+    model = _model()
+    ddp = StokeDDPModule(model, pg=pg)
+    opt = OSSOptimizer(
+        [p for p in model.parameters()], optim=torch.optim.Adam, pg=pg, lr=0.01
+    )
+    for step in range(2):
+        x, y = _data(rank * 31 + step)
+        nn.CrossEntropyLoss()(ddp(x), y).backward()
+        ddp.finish_backward()
+        opt.step()
+        opt.zero_grad()
+    io_ops.save_checkpoint(
+        runner=_R(), shard="oss", model=model, optimizer=opt,
+        path=str(tmpdir), backward_step=2, grad_accum_step=0,
+        optimizer_step=2, name="w2", status={}, verbose=False,
+    )
+    if rank == 0:
+        torch.save([p.detach().clone() for p in model.parameters()],
+                   os.path.join(str(tmpdir), "expected.pt"))
+    dist.destroy_process_group()
+
+
+def test_checkpoint_world2_loads_world1(tmp_path):
+    """OSS checkpoints are world-size independent: saved at world 2,
+    loadable by a plain single-process optimizer (reference contract,
+    SURVEY.md 5.4)."""
+    torch.multiprocessing.spawn(
+        _save_w2_worker, args=(2, free_port(), str(tmp_path)), nprocs=2,
+        join=True,
+    )
+    ckpt = torch.load(
+        os.path.join(str(tmp_path), "stoke-w2-backward-step-2.pt"),
+        weights_only=False,
+    )
+    model = _model()
+    model.load_state_dict(ckpt["model_state_dict"])
+    expected = torch.load(os.path.join(str(tmp_path), "expected.pt"),
+                          weights_only=False)
+    for p, e in zip(model.parameters(), expected):
+        assert torch.equal(p.detach(), e)
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    opt.load_state_dict(ckpt["optimizer_state_dict"])  # full, re-shardable
+    # optimizer state covers every parameter
+    assert len(ckpt["optimizer_state_dict"]["state"]) == len(expected)
+    # and training can continue
+    x, y = _data(123)
+    nn.CrossEntropyLoss()(model(x), y).backward()
+    opt.step()
+
+
+def test_benchmark_model_shapes_cpu():
+    from benchmarks.models import GPT2, Llama
+
+    g = GPT2(vocab=128, d=64, nlayer=2, nh=4, max_seq=32)
+    out = g(torch.randint(0, 128, (2, 16)))
+    assert out.shape == (2, 16, 128)
+    out.float().pow(2).mean().backward()
+
+    m = Llama(vocab=64, d=64, nlayer=2, nh=4, nkv=2, ffn=128, max_seq=32)
+    out = m(torch.randint(0, 64, (2, 16)))
+    assert out.shape == (2, 16, 64)
+    out.float().pow(2).mean().backward()
