@@ -180,3 +180,79 @@ def test_benchmark_model_shapes_cpu():
     out = m(torch.randint(0, 64, (2, 16)))
     assert out.shape == (2, 16, 64)
     out.float().pow(2).mean().backward()
+
+
+def _fsdp_save_w2_worker(rank, world, port, tmpdir):
+    pg = init_gloo(rank, world, port)
+    from stoke import io_ops
+    from stoke.shard import StokeFSDPModule
+
+    class _R:
+        rank = pg.rank
+
+        def barrier(self):
+            pg.barrier()
+
+    model = _model()
+    fsdp = StokeFSDPModule(model, pg=pg, compute_dtype=torch.float32,
+                           min_wrap_params=100)
+    opt = torch.optim.Adam(fsdp.parameters(), lr=0.01)
+    for step in range(2):
+        x, y = _data(rank * 41 + step)
+        nn.CrossEntropyLoss()(fsdp(x), y).backward()
+        fsdp.finish_backward()
+        opt.step()
+        opt.zero_grad()
+    io_ops.save_checkpoint(
+        runner=_R(), shard="fsdp", model=fsdp, optimizer=opt,
+        path=str(tmpdir), backward_step=2, grad_accum_step=0,
+        optimizer_step=2, name="fw2", status={}, verbose=False,
+    )
+    sd = fsdp.full_state_dict()  # collective: every rank participates
+    if rank == 0:
+        torch.save(sd, os.path.join(str(tmpdir), "expected_full.pt"))
+    dist.destroy_process_group()
+
+
+def _fsdp_load_w1_worker(rank, world, port, tmpdir):
+    pg = init_gloo(rank, world, port)  # world 1
+    from stoke import io_ops
+    from stoke.shard import StokeFSDPModule
+
+    class _R:
+        rank = pg.rank
+        device_id = "cpu"
+
+        def barrier(self):
+            pg.barrier()
+
+    model = _model()
+    fsdp = StokeFSDPModule(model, pg=pg, compute_dtype=torch.float32,
+                           min_wrap_params=100)
+    opt = torch.optim.Adam(fsdp.parameters(), lr=0.01)
+    io_ops.load_checkpoint(
+        runner=_R(), shard="fsdp", model=fsdp, optimizer=opt, gpu=False,
+        path=str(tmpdir), tag="stoke-fw2-backward-step-2.pt",
+    )
+    got = fsdp.full_state_dict()
+    want = torch.load(os.path.join(str(tmpdir), "expected_full.pt"),
+                      weights_only=False)
+    for k in want:
+        assert torch.allclose(got[k], want[k], atol=1e-6), k
+    # training continues from the re-sharded optimizer state
+    x, y = _data(7)
+    nn.CrossEntropyLoss()(fsdp(x), y).backward()
+    fsdp.finish_backward()
+    opt.step()
+    dist.destroy_process_group()
+
+
+def test_fsdp_checkpoint_world2_loads_world1(tmp_path):
+    torch.multiprocessing.spawn(
+        _fsdp_save_w2_worker, args=(2, free_port(), str(tmp_path)), nprocs=2,
+        join=True,
+    )
+    torch.multiprocessing.spawn(
+        _fsdp_load_w1_worker, args=(1, free_port(), str(tmp_path)), nprocs=1,
+        join=True,
+    )
