@@ -100,3 +100,44 @@ def test_bucket_overlap_residual_batches():
     _, s_no = build(n=1100, buckets=4, batch=8, replicas=2, drop_last=True,
                     allow_bucket_overlap=False)
     assert len(s) >= len(s_no)
+
+
+@pytest.mark.parametrize("n,buckets,batch,replicas,drop_last", [
+    (400, 2, 4, 2, False),
+    (407, 2, 4, 2, True),
+    (1000, 4, 8, 4, True),
+    (513, 2, 8, 2, False),
+    (640, 2, 10, 8, False),
+])
+def test_sampler_grid_coverage(n, buckets, batch, replicas, drop_last):
+    """Across a parameter grid: per-replica counts equal, per-slice indices
+    disjoint across replicas, every index within dataset bounds, epochs
+    reshuffle deterministically."""
+    ds = list(range(n))
+    sorted_idx = list(range(n))
+    samplers = [
+        BucketedDistributedSampler(
+            ds, buckets=buckets, batch_size=batch, sorted_idx=sorted_idx,
+            num_replicas=replicas, rank=r, drop_last=drop_last, seed=3,
+        )
+        for r in range(replicas)
+    ]
+    lens = {len(s) for s in samplers}
+    assert len(lens) == 1
+    per_rank = [list(iter(s)) for s in samplers]
+    for idxs in per_rank:
+        assert len(idxs) == len(samplers[0])
+        assert all(0 <= i < n for i in idxs)
+    # batch b of rank r must not overlap batch b of any other rank
+    nb = len(per_rank[0]) // batch
+    for b in range(nb):
+        seen = set()
+        for idxs in per_rank:
+            sl = set(idxs[b * batch : (b + 1) * batch])
+            assert not (sl & seen), f"overlap in slice {b}"
+            seen |= sl
+    # same epoch => same order; different epoch => different order
+    again = list(iter(samplers[0]))
+    assert again == per_rank[0]
+    samplers[0].set_epoch(1)
+    assert list(iter(samplers[0])) != per_rank[0]
