@@ -499,6 +499,11 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor x, at::Tensor dy,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor invstd);
 
+// Flash-attention forward — ROUND-2 WIP, env-gated tests only (csrc/fa_fwd.hip)
+void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D);
+std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               bool causal);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
@@ -514,6 +519,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd, "fused bf16 SwiGLU backward");
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
+  m.def("mfma_probe", &mfma_probe,
+        "16x16x32 bf16 MFMA fragment-layout probe (round-2 WIP)");
+  m.def("fa_fwd", &fa_fwd,
+        "flash-attention forward v0 (round-2 WIP; not wired into models)");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

@@ -35,7 +35,8 @@ setup(
                      "csrc/fused_rmsnorm.hip",
                      "csrc/fused_rope.hip",
                      "csrc/fused_swiglu.hip",
-                     "csrc/fused_layernorm.hip"],
+                     "csrc/fused_layernorm.hip",
+                     "csrc/fa_fwd.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],
