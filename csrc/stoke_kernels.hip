@@ -504,6 +504,12 @@ void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D);
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                bool causal);
 
+// fp8 fused quantize — ROUND-2 WIP, env-gated tests only (csrc/fp8_quant.hip)
+at::Tensor fp8_quant(at::Tensor x, at::Tensor scale, at::Tensor amax_next,
+                     int64_t kind);
+std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
+                                    at::Tensor amax_next, int64_t kind);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
@@ -523,6 +529,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe (round-2 WIP)");
   m.def("fa_fwd", &fa_fwd,
         "flash-attention forward v0 (round-2 WIP; not wired into models)");
+  m.def("fp8_quant", &fp8_quant,
+        "fused bf16->fp8 quantize + next-amax (round-2 WIP)");
+  m.def("fp8_quant_t", &fp8_quant_t,
+        "fused bf16->fp8 quantize emitting both layouts (round-2 WIP)");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,

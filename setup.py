@@ -36,7 +36,8 @@ setup(
                      "csrc/fused_rope.hip",
                      "csrc/fused_swiglu.hip",
                      "csrc/fused_layernorm.hip",
-                     "csrc/fa_fwd.hip"],
+                     "csrc/fa_fwd.hip",
+                     "csrc/fp8_quant.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],
