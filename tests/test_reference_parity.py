@@ -95,3 +95,44 @@ def test_enums_match_reference():
         rnames = [m.name for m in rcls]
         onames = [m.name for m in ocls]
         assert rnames == onames, f"{ename}: {onames} != {rnames}"
+
+
+def test_sampler_lengths_match_reference():
+    """Per-replica sample counts equal the reference's
+    BucketedDistributedSampler across a parameter grid (including the
+    bucket-overlap residual batches)."""
+    import sys
+    import types
+
+    if not os.path.exists("/root/reference/stoke/data.py"):
+        pytest.skip("reference checkout not present")
+    # the reference imports horovod unconditionally; stub it
+    if "horovod" not in sys.modules:
+        h = types.ModuleType("horovod")
+        ht = types.ModuleType("horovod.torch")
+        h.torch = ht
+        sys.modules["horovod"] = h
+        sys.modules["horovod.torch"] = ht
+    spec = importlib.util.spec_from_file_location(
+        "ref_data", "/root/reference/stoke/data.py"
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    from stoke.data import BucketedDistributedSampler as Ours
+
+    ds = list(range(1000))
+    si = list(range(1000))
+    for buckets, batch, reps, dl, overlap in [
+        (2, 4, 2, False, False),
+        (2, 4, 2, True, False),
+        (4, 8, 4, True, False),
+        (2, 8, 2, False, False),
+        (4, 8, 4, True, True),
+    ]:
+        kw = dict(buckets=buckets, batch_size=batch, sorted_idx=si,
+                  backend=None, allow_bucket_overlap=overlap,
+                  num_replicas=reps, rank=0, drop_last=dl)
+        r = mod.BucketedDistributedSampler(ds, **kw)
+        o = Ours(ds, **kw)
+        assert len(r) == len(o), (buckets, batch, reps, dl, overlap)
+        assert len(list(iter(r))) == len(list(iter(o)))
