@@ -503,6 +503,9 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor x, at::Tensor dy,
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D);
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                bool causal);
+std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               at::Tensor out, at::Tensor dout,
+                               at::Tensor lse, bool causal);
 
 // fp8 fused quantize — ROUND-2 WIP, env-gated tests only (csrc/fp8_quant.hip)
 at::Tensor fp8_quant(at::Tensor x, at::Tensor scale, at::Tensor amax_next,
@@ -529,6 +532,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe (round-2 WIP)");
   m.def("fa_fwd", &fa_fwd,
         "flash-attention forward v0 (round-2 WIP; not wired into models)");
+  m.def("fa_bwd", &fa_bwd,
+        "flash-attention backward v0 (round-2 WIP; not wired into models)");
   m.def("fp8_quant", &fp8_quant,
         "fused bf16->fp8 quantize + next-amax (round-2 WIP)");
   m.def("fp8_quant_t", &fp8_quant_t,

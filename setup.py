@@ -37,6 +37,7 @@ setup(
                      "csrc/fused_swiglu.hip",
                      "csrc/fused_layernorm.hip",
                      "csrc/fa_fwd.hip",
+                     "csrc/fa_bwd.hip",
                      "csrc/fp8_quant.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],

@@ -81,3 +81,33 @@ def test_fa_fwd_vs_sdpa(B, H, HKV, S, D, causal):
     want_lse = torch.logsumexp(s, dim=-1)
     lerr = (lse - want_lse).abs().max().item()
     assert lerr < 0.05, f"lse err {lerr}"
+
+
+@pytest.mark.parametrize("B,H,HKV,S,D,causal", [
+    (1, 2, 2, 64, 64, True),
+    (1, 2, 1, 128, 128, True),
+    (1, 1, 1, 96, 64, False),
+])
+def test_fa_bwd_vs_sdpa(B, H, HKV, S, D, causal):
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from stoke.nn.attention import flash_attention
+
+    torch.manual_seed(2)
+    q = torch.randn(B, H, S, D, device="cuda").bfloat16().requires_grad_(True)
+    k = torch.randn(B, HKV, S, D, device="cuda").bfloat16().requires_grad_(True)
+    v = torch.randn(B, HKV, S, D, device="cuda").bfloat16().requires_grad_(True)
+    gy = torch.randn(B, H, S, D, device="cuda")
+    flash_attention(q, k, v, causal=causal).backward(gy.bfloat16())
+
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    torch.nn.functional.scaled_dot_product_attention(
+        q32, k32, v32, is_causal=causal, enable_gqa=(H != HKV)
+    ).backward(gy)
+    for got, want, nm in ((q.grad, q32.grad, "dq"), (k.grad, k32.grad, "dk"),
+                          (v.grad, v32.grad, "dv")):
+        err = (got.float() - want).abs().max().item()
+        scale = want.abs().max().item() + 1e-6
+        assert err / scale < 0.06, f"{nm} err {err} (scale {scale})"
