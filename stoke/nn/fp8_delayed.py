@@ -1,0 +1,111 @@
+# -*- coding: utf-8 -*-
+"""Delayed-scaling FP8 linear — ROUND-2 WORK IN PROGRESS.
+
+Rebuilds FP8Linear's GEMM path on the fused quantize kernels
+(csrc/fp8_quant.hip): scales come from the PREVIOUS step's amax (no extra
+reduction pass — the next amax is a byproduct of the cast), and the
+backward's column-major operands come from the dual-layout quantizer (no
+``.t().contiguous()`` copies).  This removes every v1 overhead the llama
+profile identified (NOTES.md).
+
+UNVALIDATED on hardware as of round 1; nothing uses this class by default.
+Round 2: STOKE_FP8V2_TEST=1 validates the kernels, then this module, then
+benchmark `--fp8` can switch to it.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+_E4M3_MAX = 448.0
+_E5M2_MAX = 57344.0
+_MARGIN = 2.0  # headroom factor on the delayed amax
+
+
+class _FP8DelayedMMFn(torch.autograd.Function):
+    """y = x @ w^T with delayed per-tensor scales held in module buffers."""
+
+    @staticmethod
+    def forward(ctx, x, w, sx, sw, sdy, ax, aw, ady):
+        from stoke import _C
+
+        # quantize with LAST step's scales; accumulate this step's amax
+        ax.zero_()
+        aw.zero_()
+        x8, x8t = _C.fp8_quant_t(x, sx, ax, 0)
+        w8, w8t = _C.fp8_quant_t(w, sw, aw, 0)
+        y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
+                             out_dtype=torch.bfloat16)
+        ctx.save_for_backward(x8t, w8t, sx, sw, sdy, ady)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from stoke import _C
+
+        x8t, w8t, sx, sw, sdy, ady = ctx.saved_tensors
+        ady.zero_()
+        dy8, dy8t = _C.fp8_quant_t(dy.contiguous(), sdy, ady, 1)
+        # dx[T,K] = dy[T,N] @ w[N,K]; w col-major [N,K] is w8t[K,N].t()
+        dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sdy, scale_b=sw,
+                              out_dtype=torch.bfloat16)
+        # dw[N,K] = dy^T[N,T] @ x[T,K]; x col-major [T,K] is x8t[K,T].t()
+        dw = torch._scaled_mm(dy8t, x8t.t(), scale_a=sdy, scale_b=sx,
+                              out_dtype=torch.bfloat16)
+        return dx, dw, None, None, None, None, None, None
+
+
+class FP8LinearDelayed(nn.Linear):
+    """nn.Linear with delayed-scaling fp8 GEMMs (round-2 WIP)."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = True,
+                 device=None, dtype=None):
+        super().__init__(in_features, out_features, bias=bias, device=device,
+                         dtype=dtype)
+        for name in ("_sx", "_sw", "_sdy"):
+            self.register_buffer(name, torch.ones(1), persistent=False)
+        for name in ("_ax", "_aw", "_ady"):
+            self.register_buffer(name, torch.zeros(1), persistent=False)
+        self._primed = False
+
+    def _update_scales(self):
+        # next step's scale from this step's amax (device-side, no sync)
+        torch.clamp(self._ax * (_MARGIN / _E4M3_MAX), min=1e-12,
+                    out=self._sx)
+        torch.clamp(self._aw * (_MARGIN / _E4M3_MAX), min=1e-12,
+                    out=self._sw)
+        torch.clamp(self._ady * (_MARGIN / _E5M2_MAX), min=1e-12,
+                    out=self._sdy)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from stoke import ops
+        from stoke.nn.fp8 import fp8_available
+
+        usable = (
+            x.is_cuda and x.dtype == torch.bfloat16
+            and self.weight.dtype == torch.bfloat16
+            and x.shape[-1] % 16 == 0 and self.out_features % 16 == 0
+            and ops.has_ext() and fp8_available()
+        )
+        if not usable:
+            return super().forward(x)
+        if self._primed:
+            self._update_scales()
+        shape = x.shape[:-1] + (self.out_features,)
+        flat = x.reshape(-1, x.shape[-1]).contiguous()
+        if flat.shape[0] % 16 != 0:
+            return super().forward(x)
+        if not self._primed:
+            # first step: one-off direct amax so scales start sane
+            self._ax.copy_(flat.float().abs().amax().reshape(1))
+            self._aw.copy_(self.weight.float().abs().amax().reshape(1))
+            self._ady.fill_(1.0)
+            self._update_scales()
+            self._primed = True
+        y = _FP8DelayedMMFn.apply(flat, self.weight, self._sx, self._sw,
+                                  self._sdy, self._ax, self._aw, self._ady)
+        y = y.reshape(shape)
+        if self.bias is not None:
+            y = y + self.bias
+        return y

@@ -60,3 +60,26 @@ def test_fp8_quant_t_layouts():
         yt.view(torch.uint8), want.t().contiguous().view(torch.uint8)
     )
     assert abs(amax.item() - x.float().abs().max().item()) < 1e-2
+
+
+def test_fp8_delayed_linear_numerics():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from stoke.nn.fp8 import fp8_available
+    from stoke.nn.fp8_delayed import FP8LinearDelayed
+
+    if not fp8_available():
+        pytest.skip("fp8 unavailable")
+    torch.manual_seed(3)
+    lin = FP8LinearDelayed(256, 512).cuda().bfloat16()
+    x = torch.randn(64, 256, device="cuda").bfloat16().requires_grad_(True)
+    # two steps: step 1 primes scales, step 2 runs fully delayed
+    for _ in range(2):
+        y = lin(x)
+        y.float().pow(2).mean().backward()
+    ref = torch.nn.functional.linear(x.float(), lin.weight.float(),
+                                     lin.bias.float())
+    rel = (y.float() - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.1, f"fp8 delayed fwd rel err {rel}"
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    assert lin.weight.grad is not None
