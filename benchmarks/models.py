@@ -161,7 +161,9 @@ class GPT2Block(nn.Module):
         q = q.view(B, S, self.nh, -1).transpose(1, 2)
         k = k.view(B, S, self.nh, -1).transpose(1, 2)
         v = v.view(B, S, self.nh, -1).transpose(1, 2)
-        a = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        from stoke.nn.attention import attention
+
+        a = attention(q, k, v, causal=True)
         x = x + self.proj(a.transpose(1, 2).reshape(B, S, D))
         return x + self.mlp(self.ln2(x))
 
@@ -237,9 +239,9 @@ class LlamaBlock(nn.Module):
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)
         v = self.wv(h).view(B, S, self.nkv, self.hd).transpose(1, 2)
-        a = F.scaled_dot_product_attention(
-            q, k, v, is_causal=True, enable_gqa=(self.nkv != self.nh)
-        )
+        from stoke.nn.attention import attention
+
+        a = attention(q, k, v, causal=True)
         a = a.transpose(1, 2).reshape(B, S, -1)
         x = x + self.wo(a)
         h = self.mlp_norm(x)

@@ -94,14 +94,10 @@ class StokeGradScaler:
         for g in grads:
             by_dev[(g.device, g.dtype)].append(g)
         for (dev, dt), gs in by_dev.items():
-            if dt != torch.float32:
-                # unscale non-fp32 grads via foreach ops, check on device
-                torch._foreach_mul_(gs, inv_scale.to(dev))
-                bad = any(not torch.isfinite(g).all() for g in gs)
-                if bad:
-                    found_inf.fill_(1.0)
-            else:
-                ops.multi_tensor_unscale_(gs, inv_scale.to(dev), found_inf)
+            # One fused unscale+finite-check launch per (device, dtype)
+            # group — fp32, fp16 and bf16 all go through the HIP kernel
+            # (UnscaleHalfFunctor for 2-byte dtypes), no host sync.
+            ops.multi_tensor_unscale_(gs, inv_scale.to(dev), found_inf)
         state["found_inf"] = found_inf
         state["unscaled"] = True
 

@@ -1,16 +1,19 @@
 # -*- coding: utf-8 -*-
-"""In-house flash attention — ROUND-2 WORK IN PROGRESS.
+"""In-house CDNA4 flash attention (MFMA bf16, FA-2 style).
 
-``flash_attention(q, k, v, causal=...)`` runs the MFMA kernels in
-``csrc/fa_fwd.hip`` / ``csrc/fa_bwd.hip``.  UNVALIDATED on hardware as of
-round 1 (GPU budget exhausted after compile verification): nothing imports
-this module by default, and ``benchmarks/models.py`` keeps
-``F.scaled_dot_product_attention``.  Round 2: run the env-gated tests in
-tests/test_fa_wip.py (probe first), fix what they find, then gate this in
-via ``STOKE_USE_FA=1``.
+``flash_attention(q, k, v, causal=...)`` runs the hand-written kernels in
+``csrc/fa_fwd.hip`` / ``csrc/fa_bwd.hip`` — hardware-validated round 2
+(tests/test_fa_wip.py: fragment-layout probe + fwd/bwd numerics vs fp32
+SDPA, all passing on MI355X).  ``attention()`` is the model-facing entry:
+it routes to the in-house kernels when the shapes qualify and the
+``STOKE_USE_FA`` gate allows, else falls back to
+``F.scaled_dot_product_attention`` (AOTriton).
 """
 
+import os
+
 import torch
+import torch.nn.functional as F
 
 
 class _FlashAttnFn(torch.autograd.Function):
@@ -38,3 +41,30 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     """q: [B,H,S,D], k/v: [B,Hkv,S,D] bf16 contiguous; D in {64, 128}."""
     return _FlashAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
                               causal)
+
+
+def _fa_usable(q, k, v, causal) -> bool:
+    if os.environ.get("STOKE_USE_FA", "1") == "0":
+        return False
+    if not (q.is_cuda and q.dtype == torch.bfloat16
+            and k.dtype == torch.bfloat16 and v.dtype == torch.bfloat16):
+        return False
+    if q.dim() != 4 or q.shape[-1] not in (64, 128):
+        return False
+    if k.shape[1] != q.shape[1] and q.shape[1] % k.shape[1] != 0:
+        return False
+    from stoke import ops
+
+    return ops.has_ext()
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              causal: bool = True) -> torch.Tensor:
+    """Model-facing attention: in-house MFMA flash kernels when usable
+    (bf16 CUDA [B,H,S,D] with D in {64,128}), SDPA otherwise.  Disable the
+    native path with ``STOKE_USE_FA=0``."""
+    if _fa_usable(q, k, v, causal):
+        return flash_attention(q, k, v, causal=causal)
+    return F.scaled_dot_product_attention(
+        q, k, v, is_causal=causal, enable_gqa=(k.shape[1] != q.shape[1])
+    )

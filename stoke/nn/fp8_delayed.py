@@ -37,14 +37,24 @@ class _FP8DelayedMMFn(torch.autograd.Function):
         w8, w8t = _C.fp8_quant_t(w, sw, aw, 0)
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              out_dtype=torch.bfloat16)
-        ctx.save_for_backward(x8t, w8t, sx, sw, sdy, ady)
+        # Scales are saved as CLONES: _update_scales() mutates the module
+        # buffers in place between micro-steps, and backward must see the
+        # forward-time values (also avoids the version-counter trip under
+        # gradient accumulation — ADVICE.md round 1).  ady is shared on
+        # purpose (backward writes the next step's dy amax into it) and
+        # lives on ctx, outside version tracking.
+        ctx.save_for_backward(x8t, w8t)
+        ctx.fp8_scales = (sx.clone(), sw.clone(), sdy.clone())
+        ctx.fp8_ady = ady
         return y
 
     @staticmethod
     def backward(ctx, dy):
         from stoke import _C
 
-        x8t, w8t, sx, sw, sdy, ady = ctx.saved_tensors
+        x8t, w8t = ctx.saved_tensors
+        sx, sw, sdy = ctx.fp8_scales
+        ady = ctx.fp8_ady
         ady.zero_()
         dy8, dy8t = _C.fp8_quant_t(dy.contiguous(), sdy, ady, 1)
         # dx[T,K] = dy[T,N] @ w[N,K]; w col-major [N,K] is w8t[K,N].t()
@@ -69,6 +79,7 @@ class FP8LinearDelayed(nn.Linear):
             self.register_buffer(name, torch.zeros(1), persistent=False)
         self._primed = False
 
+    @torch.no_grad()
     def _update_scales(self):
         # next step's scale from this step's amax (device-side, no sync)
         torch.clamp(self._ax * (_MARGIN / _E4M3_MAX), min=1e-12,
@@ -97,10 +108,13 @@ class FP8LinearDelayed(nn.Linear):
         if flat.shape[0] % 16 != 0:
             return super().forward(x)
         if not self._primed:
-            # first step: one-off direct amax so scales start sane
-            self._ax.copy_(flat.float().abs().amax().reshape(1))
-            self._aw.copy_(self.weight.float().abs().amax().reshape(1))
-            self._ady.fill_(1.0)
+            # first step: one-off direct amax so scales start sane.
+            # no_grad: copying a grad-requiring amax into the buffer would
+            # silently flip the buffer's requires_grad (round-2 GPU finding).
+            with torch.no_grad():
+                self._ax.copy_(flat.float().abs().amax().reshape(1))
+                self._aw.copy_(self.weight.float().abs().amax().reshape(1))
+                self._ady.fill_(1.0)
             self._update_scales()
             self._primed = True
         y = _FP8DelayedMMFn.apply(flat, self.weight, self._sx, self._sw,

@@ -154,6 +154,61 @@ def fused_adamw_(
             p.copy_(w.to(p.dtype))
 
 
+def fused_sgd_(
+    params: List[torch.Tensor],
+    grads: List[torch.Tensor],
+    momentum_bufs: List[torch.Tensor],
+    lr: float,
+    momentum: float = 0.0,
+    dampening: float = 0.0,
+    weight_decay: float = 0.0,
+    nesterov: bool = False,
+    first_step: bool = False,
+    found_inf: Optional[torch.Tensor] = None,
+    inv_scale: Optional[torch.Tensor] = None,
+    masters: Optional[List[torch.Tensor]] = None,
+):
+    """Single-launch fused SGD(+momentum) over the whole parameter list.
+
+    ``torch.optim.SGD`` semantics including the first-step ``buf = g``
+    initialization.  fp32 path, or bf16 params+grads with fp32
+    masters+buffers when ``masters`` is given.
+    """
+    if not params:
+        return
+    if params[0].is_cuda:
+        ext = _require_ext()
+        if masters is not None:
+            ext.multi_tensor_sgd_bf16_(
+                params, grads, momentum_bufs, masters, lr, momentum,
+                dampening, weight_decay, nesterov, first_step,
+                found_inf, inv_scale,
+            )
+        else:
+            ext.multi_tensor_sgd_(
+                params, grads, momentum_bufs, lr, momentum, dampening,
+                weight_decay, nesterov, first_step, found_inf, inv_scale,
+            )
+        return
+    # CPU reference implementation (the numerics oracle for GPU tests)
+    if found_inf is not None and found_inf.item() != 0:
+        return
+    inv = inv_scale.item() if inv_scale is not None else 1.0
+    for i, p in enumerate(params):
+        w = masters[i] if masters is not None else p
+        g = grads[i].float() * inv + weight_decay * w.float()
+        if momentum != 0.0:
+            b = momentum_bufs[i]
+            if first_step:
+                b.copy_(g)
+            else:
+                b.mul_(momentum).add_(g, alpha=1 - dampening)
+            g = g + momentum * b if nesterov else b.clone()
+        w.add_(g, alpha=-lr)
+        if masters is not None:
+            p.copy_(w.to(p.dtype))
+
+
 def amp_update_scale_(
     scale: torch.Tensor,
     growth_tracker: torch.Tensor,

@@ -64,6 +64,35 @@ def test_unscale_misaligned_view():
     assert found.item() == 0.0
 
 
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_unscale_half_dtypes(dtype):
+    """fp16/bf16 grads go through the fused kernel — no host-sync fallback."""
+    from stoke import ops
+
+    g1 = (torch.randn(1 << 18, device="cuda") * 4).to(dtype)
+    g2 = (torch.randn(517, device="cuda") * 4).to(dtype)
+    ref1, ref2 = g1.float(), g2.float()
+    inv = torch.tensor([0.25], device="cuda")
+    found = torch.zeros(1, device="cuda")
+    ops.multi_tensor_unscale_([g1, g2], inv, found)
+    torch.cuda.synchronize()
+    assert found.item() == 0.0
+    assert torch.allclose(g1.float(), (ref1 * 0.25).to(dtype).float())
+    assert torch.allclose(g2.float(), (ref2 * 0.25).to(dtype).float())
+    # misaligned view exercises the scalar tail
+    base = (torch.randn((1 << 12) + 9, device="cuda")).to(dtype)
+    view = base[3:]
+    refv = view.float()
+    ops.multi_tensor_unscale_([view], inv, found)
+    torch.cuda.synchronize()
+    assert torch.allclose(view.float(), (refv * 0.25).to(dtype).float())
+    # inf detection
+    g2[7] = float("inf")
+    ops.multi_tensor_unscale_([g1, g2], inv, found)
+    torch.cuda.synchronize()
+    assert found.item() == 1.0
+
+
 def test_l2norm_vs_torch():
     from stoke import ops
 
@@ -127,6 +156,63 @@ def test_fused_adamw_vs_torch_fp32():
     for pa, pb in zip(p_a, p_b):
         err = (pa - pb).abs().max().item()
         assert err < 1e-5, f"fused adamw deviates: {err}"
+
+
+@pytest.mark.parametrize("momentum,dampening,nesterov,wd", [
+    (0.0, 0.0, False, 0.0),
+    (0.9, 0.0, False, 1e-4),
+    (0.9, 0.1, False, 0.0),
+    (0.9, 0.0, True, 1e-4),
+])
+def test_fused_sgd_vs_torch_fp32(momentum, dampening, nesterov, wd):
+    """HIP fused SGD against torch.optim.SGD on identical fp32 inputs."""
+    from stoke.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(0)
+    shapes = [(1 << 16,), (513,), (33, 77), (3,)]
+    init = [torch.randn(s, device="cuda") for s in shapes]
+    p_a = [torch.nn.Parameter(t.clone()) for t in init]
+    p_b = [torch.nn.Parameter(t.clone()) for t in init]
+    opt_a = FusedSGD(p_a, lr=0.05, momentum=momentum, dampening=dampening,
+                     nesterov=nesterov, weight_decay=wd)
+    opt_b = torch.optim.SGD(p_b, lr=0.05, momentum=momentum,
+                            dampening=dampening, nesterov=nesterov,
+                            weight_decay=wd)
+    for step in range(8):
+        torch.manual_seed(step)
+        gs = [torch.randn_like(t) for t in init]
+        for pa, pb, g in zip(p_a, p_b, gs):
+            pa.grad = g.clone()
+            pb.grad = g.clone()
+        opt_a.step()
+        opt_b.step()
+    torch.cuda.synchronize()
+    for pa, pb in zip(p_a, p_b):
+        err = (pa - pb).abs().max().item()
+        assert err < 1e-5, f"fused sgd deviates: {err}"
+
+
+def test_fused_sgd_bf16_master():
+    """bf16 param + fp32 master SGD path vs fp32 torch.optim.SGD oracle."""
+    from stoke.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(0)
+    init = torch.randn(1 << 14, device="cuda")
+    p32 = torch.nn.Parameter(init.clone())
+    p16 = torch.nn.Parameter(init.clone().bfloat16())
+    o32 = torch.optim.SGD([p32], lr=0.05, momentum=0.9)
+    o16 = FusedSGD([p16], lr=0.05, momentum=0.9)
+    for step in range(5):
+        torch.manual_seed(step)
+        g = torch.randn_like(init)
+        p32.grad = g.clone()
+        p16.grad = g.bfloat16()
+        o32.step()
+        o16.step()
+    torch.cuda.synchronize()
+    master = o16.state[p16]["master"]
+    assert torch.allclose(master, p32.detach(), rtol=3e-2, atol=3e-3)
+    assert torch.equal(p16.detach(), master.to(torch.bfloat16))
 
 
 def test_fused_adamw_bf16_master():

@@ -134,3 +134,68 @@ def test_scaler_step_with_fused_optimizer_device_flag():
     sc.step(opt)   # goes through found_inf kwarg path
     sc.update()
     assert not torch.equal(p.detach(), torch.ones(8))
+
+
+# ------------------------------------------------------------- FusedSGD
+def test_fused_sgd_matches_torch_sgd():
+    from stoke.ops.fused_sgd import FusedSGD
+
+    for momentum, dampening, nesterov, wd in [
+        (0.0, 0.0, False, 0.0),
+        (0.9, 0.0, False, 1e-4),
+        (0.9, 0.1, False, 0.0),
+        (0.9, 0.0, True, 1e-4),
+    ]:
+        torch.manual_seed(0)
+        shapes = [(37,), (8, 9), (4, 5, 6)]
+        init = [torch.randn(s) for s in shapes]
+        p_a = [torch.nn.Parameter(t.clone()) for t in init]
+        p_b = [torch.nn.Parameter(t.clone()) for t in init]
+        opt_a = FusedSGD(p_a, lr=0.05, momentum=momentum,
+                         dampening=dampening, nesterov=nesterov,
+                         weight_decay=wd)
+        opt_b = torch.optim.SGD(p_b, lr=0.05, momentum=momentum,
+                                dampening=dampening, nesterov=nesterov,
+                                weight_decay=wd)
+        for step in range(5):
+            torch.manual_seed(100 + step)
+            grads = [torch.randn_like(t) for t in init]
+            for pa, pb, g in zip(p_a, p_b, grads):
+                pa.grad = g.clone()
+                pb.grad = g.clone()
+            opt_a.step()
+            opt_b.step()
+        for pa, pb in zip(p_a, p_b):
+            assert torch.allclose(pa, pb, rtol=1e-5, atol=1e-7), \
+                (momentum, dampening, nesterov, wd, (pa - pb).abs().max())
+
+
+def test_fused_sgd_bf16_master_path():
+    from stoke.ops.fused_sgd import FusedSGD
+
+    torch.manual_seed(0)
+    p32 = torch.nn.Parameter(torch.randn(64))
+    p16 = torch.nn.Parameter(p32.detach().to(torch.bfloat16))
+    o32 = torch.optim.SGD([p32], lr=0.05, momentum=0.9)
+    o16 = FusedSGD([p16], lr=0.05, momentum=0.9)
+    for step in range(3):
+        g = torch.randn(64)
+        p32.grad = g.clone()
+        p16.grad = g.to(torch.bfloat16)
+        o32.step()
+        o16.step()
+    master = o16.state[p16]["master"]
+    assert torch.allclose(master, p32.detach(), rtol=3e-2, atol=3e-3)
+    assert torch.equal(p16.detach(), master.to(torch.bfloat16))
+
+
+def test_fused_sgd_scaler_skip_on_inf():
+    from stoke.ops.fused_sgd import FusedSGD
+
+    p = torch.nn.Parameter(torch.ones(8))
+    opt = FusedSGD([p], lr=0.1, momentum=0.9)
+    p.grad = torch.ones(8)
+    opt.step(found_inf=torch.ones(1))  # must skip
+    assert torch.equal(p.detach(), torch.ones(8))
+    opt.step(found_inf=torch.zeros(1))
+    assert not torch.equal(p.detach(), torch.ones(8))
