@@ -7,7 +7,11 @@ Llama-3-8B) so the STOKE_USE_FA default is set from measurement, not hope.
 Run on the GPU box:  python benchmarks/fa_bench.py
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F

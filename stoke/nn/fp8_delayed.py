@@ -101,6 +101,11 @@ class FP8LinearDelayed(nn.Linear):
         )
         if not usable:
             return super().forward(x)
+        if self._sx.dtype != torch.float32:
+            # module-level .bfloat16()/.half() converts registered buffers;
+            # the quantizer contract is fp32 scales/amax — restore them
+            for name in ("_sx", "_sw", "_sdy", "_ax", "_aw", "_ady"):
+                setattr(self, name, getattr(self, name).float())
         if self._primed:
             self._update_scales()
         shape = x.shape[:-1] + (self.out_features,)

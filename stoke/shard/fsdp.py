@@ -99,31 +99,72 @@ class _ShardUnit:
         for p, o in zip(self.params, self.offsets):
             p.data = self.full_flat[o : o + p.numel()].view(p._orig_shape)
         self.materialized = True
+        self._gather_event = None
+        self._gather_pending = False
+        self._rs_event = None
+        self._pending_shard_g = None
         self.free_full()
         self.grad_ready = 0
         self.rs_done = False
 
     # ------------------------------------------------------------- gather
-    def gather(self):
+    def gather(self, comm_stream=None):
+        """Materialize full params via RCCL all-gather.
+
+        With ``comm_stream`` the collective runs on a side stream so the
+        NEXT unit's gather overlaps the CURRENT unit's compute (forward and
+        backward prefetch — SURVEY.md section 7 hard-part #1).  Consumers
+        must call :meth:`wait_gather` on the compute stream first.
+        """
         if self.materialized:
             return
+        # Storage (re)allocated on the compute stream: the allocator then
+        # owns the block on the stream that also frees it (free_full), and
+        # the comm stream's writes are fenced by record_stream below.
         self.full_flat.untyped_storage().resize_(self._full_bytes)
+        self.materialized = True
+        if comm_stream is not None:
+            comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(comm_stream):
+                comm_shard = self.shard.data.to(self.compute_dtype)
+                if self._pg.world_size > 1:
+                    self._pg.all_gather_flat(self.full_flat, comm_shard)
+                else:
+                    self.full_flat.copy_(comm_shard)
+                self.full_flat.record_stream(comm_stream)
+            if self._gather_event is None:
+                self._gather_event = torch.cuda.Event()
+            self._gather_event.record(comm_stream)
+            self._gather_pending = True
+            return
         comm_shard = self.shard.data.to(self.compute_dtype)
         if self._pg.world_size > 1:
             self._pg.all_gather_flat(self.full_flat, comm_shard)
         else:
             self.full_flat.copy_(comm_shard)
-        self.materialized = True
+
+    def wait_gather(self):
+        """Compute-stream fence for an async :meth:`gather`."""
+        if self._gather_pending:
+            torch.cuda.current_stream().wait_event(self._gather_event)
+            self._gather_pending = False
 
     def free_full(self):
         if not self.materialized:
             return
+        self.wait_gather()
         self.full_flat.untyped_storage().resize_(0)
         self.materialized = False
 
     # ------------------------------------------------------- grad handling
-    def reduce_scatter_grads(self, average: bool = True):
-        """Pack full grads -> one reduce-scatter -> fp32 shard gradient."""
+    def reduce_scatter_grads(self, average: bool = True, comm_stream=None):
+        """Pack full grads -> one reduce-scatter -> fp32 shard gradient.
+
+        Async variant (``comm_stream`` given): the pack runs on the compute
+        stream, the reduce-scatter on the side stream (overlapping the rest
+        of backward), and :meth:`finalize_grad` converts/accumulates after
+        an event wait.
+        """
         if self.rs_done:
             return
         self.rs_done = True
@@ -138,12 +179,46 @@ class _ShardUnit:
                     p.grad.reshape(-1).to(rs_dtype)
                 )
             p.grad = None
+        if comm_stream is not None:
+            comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(comm_stream):
+                shard_g = torch.empty(
+                    self.shard_nelem, dtype=rs_dtype, device=device
+                )
+                if self._pg.world_size > 1:
+                    self._pg.reduce_scatter_flat(shard_g, flat_g)
+                else:
+                    shard_g.copy_(flat_g)
+                flat_g.record_stream(comm_stream)
+            if self._rs_event is None:
+                self._rs_event = torch.cuda.Event()
+            self._rs_event.record(comm_stream)
+            self._pending_shard_g = shard_g
+            self.grad_ready = 0
+            self.free_full()
+            return
         shard_g = torch.empty(self.shard_nelem, dtype=rs_dtype, device=device)
         if self._pg.world_size > 1:
             self._pg.reduce_scatter_flat(shard_g, flat_g)
         else:
             shard_g.copy_(flat_g)
         del flat_g
+        self._pending_shard_g = shard_g
+        self.grad_ready = 0
+        self.free_full()
+        self.finalize_grad(average)
+
+    def finalize_grad(self, average: bool = True):
+        """Convert/accumulate the reduce-scattered gradient on the compute
+        stream (after waiting on the side-stream event if one is pending)."""
+        shard_g = self._pending_shard_g
+        if shard_g is None:
+            return
+        self._pending_shard_g = None
+        if self._rs_event is not None:
+            torch.cuda.current_stream().wait_event(self._rs_event)
+            # last read of the comm-stream product happens on compute below
+            shard_g.record_stream(torch.cuda.current_stream())
         if average and self._pg.world_size > 1:
             shard_g = shard_g.float() / self._pg.world_size
         else:
@@ -152,8 +227,6 @@ class _ShardUnit:
             self.shard.grad = shard_g
         else:
             self.shard.grad.add_(shard_g)
-        self.grad_ready = 0
-        self.free_full()
 
 
 class StokeFSDPModule(nn.Module):
@@ -209,6 +282,16 @@ class StokeFSDPModule(nn.Module):
                 self._unit_of_param[id(p)] = u
         self._register_hooks()
         self._callback_queued = False
+        # Side stream for all-gather prefetch and async reduce-scatter
+        # (RCCL only; gloo/CPU stays synchronous).  Unit execution order is
+        # recorded on the first forward and drives depth-1 prefetch after.
+        self._comm_stream = (
+            torch.cuda.Stream()
+            if pg.backend == "nccl" and torch.cuda.is_available()
+            else None
+        )
+        self._fwd_order: List[_ShardUnit] = []
+        self._order_final = False
 
     def _auto_wrap(self, root, prefix, assigned, min_params, fp32_rs):
         """Create a shard unit per sufficiently large submodule."""
@@ -248,9 +331,26 @@ class StokeFSDPModule(nn.Module):
             for p in u.params:
                 p.register_post_accumulate_grad_hook(self._grad_hook)
 
+    def _prefetch(self, u, order):
+        """Async-gather the unit AFTER ``u`` in ``order`` (depth-1 prefetch)."""
+        if self._comm_stream is None or not self._order_final:
+            return
+        try:
+            i = order.index(u)
+        except ValueError:
+            return
+        for nxt in order[i + 1:]:
+            if not nxt.materialized:
+                nxt.gather(comm_stream=self._comm_stream)
+                return
+
     def _make_pre_fwd(self, u):
         def hook(mod, args):
-            u.gather()
+            if not self._order_final and u not in self._fwd_order:
+                self._fwd_order.append(u)
+            u.gather()  # no-op if already prefetched
+            u.wait_gather()
+            self._prefetch(u, self._fwd_order)
         return hook
 
     def _make_post_fwd(self, u):
@@ -263,8 +363,15 @@ class StokeFSDPModule(nn.Module):
 
     def _make_pre_bwd(self, u):
         def hook(mod, grad_output):
-            u.gather()
+            u.gather()  # no-op if already prefetched
+            u.wait_gather()
+            # Backward visits units in reverse forward order
+            self._prefetch(u, self._bwd_order)
         return hook
+
+    @property
+    def _bwd_order(self):
+        return list(reversed(self._fwd_order))
 
     def _grad_hook(self, param):
         u = self._unit_of_param.get(id(param))
@@ -279,14 +386,15 @@ class StokeFSDPModule(nn.Module):
             self._callback_queued = True
         u.grad_ready += 1
         if u.grad_ready == len(u.params):
-            u.reduce_scatter_grads()
+            u.reduce_scatter_grads(comm_stream=self._comm_stream)
 
     def _finalize_backward(self):
         self._callback_queued = False
         for u in self.units:
             if not u.rs_done and any(p.grad is not None for p in u.params):
-                u.reduce_scatter_grads()
+                u.reduce_scatter_grads(comm_stream=self._comm_stream)
         for u in self.units:
+            u.finalize_grad()
             u.rs_done = False
 
     def finish_backward(self):
@@ -299,8 +407,11 @@ class StokeFSDPModule(nn.Module):
             for u in self.units:
                 if any(p.grad is not None for p in u.params):
                     u.rs_done = False
-                    u.reduce_scatter_grads()
+                    u.reduce_scatter_grads(comm_stream=self._comm_stream)
             self._callback_queued = False
+        for u in self.units:
+            u.finalize_grad()
+            u.rs_done = False
 
     @contextmanager
     def no_sync(self):
@@ -316,7 +427,15 @@ class StokeFSDPModule(nn.Module):
         for u in self.units:
             if u.name == "(root)":
                 u.gather()
+        # Kick off the first wrapped unit's gather on the side stream so it
+        # overlaps the root gather / input embedding work.
+        if self._order_final and self._comm_stream is not None:
+            for u in self._fwd_order:
+                if not u.materialized:
+                    u.gather(comm_stream=self._comm_stream)
+                    break
         out = self.module(*args, **kwargs)
+        self._order_final = True
         for u in self.units:
             if u.name == "(root)" and u.reshard_after_forward and not torch.is_grad_enabled():
                 u.free_full()
@@ -402,8 +521,16 @@ class StokeFSDPModule(nn.Module):
             if name in sd:
                 buf.data.copy_(sd[name].to(buf.device, buf.dtype))
 
-    def gather_full_optim_state_dict(self, optimizer) -> Optional[dict]:
-        """Full per-original-param optimizer state (rank 0 returns it)."""
+    def gather_full_optim_state_dict(self, optimizer,
+                                     rank0_only: bool = True) -> Optional[dict]:
+        """Full per-original-param optimizer state.
+
+        All ranks participate in the all-gathers; only rank 0 materializes
+        the host-side dict (others return None) unless ``rank0_only=False``
+        — at 8 ranks x 8B params the old every-rank materialization was a
+        host-memory spike.
+        """
+        emit = (not rank0_only) or self._pg.rank == 0 or self._pg.world_size == 1
         full = {"state": {}, "param_groups": []}
         shard_index = {id(u.shard): i for i, u in enumerate(self.units)}
         pidx = 0
@@ -434,26 +561,30 @@ class StokeFSDPModule(nn.Module):
                     else:
                         gathered[k] = v
                 for p, name, o in zip(u.params, u.param_names, u.offsets):
-                    entry = {}
-                    for k, v in gathered.items():
-                        if isinstance(v, torch.Tensor) and v.numel() == u.padded:
-                            entry[k] = (
-                                v[o : o + p.numel()].view(p._orig_shape).cpu().clone()
-                            )
-                        else:
-                            entry[k] = v
-                    full["state"][name_index[name]] = entry
+                    if emit:
+                        entry = {}
+                        for k, v in gathered.items():
+                            if isinstance(v, torch.Tensor) and v.numel() == u.padded:
+                                entry[k] = (
+                                    v[o : o + p.numel()].view(p._orig_shape).cpu().clone()
+                                )
+                            else:
+                                entry[k] = v
+                        full["state"][name_index[name]] = entry
                     gidx.append(name_index[name])
                 del gathered
             full["param_groups"].append(
                 {**{k: v for k, v in group.items() if k != "params"},
                  "params": gidx}
             )
-        return full if self._pg.rank == 0 or self._pg.world_size == 1 else full
+        return full if emit else None
 
     def load_full_optim_state_dict(self, optimizer, full: dict):
         """Re-shard a consolidated optimizer state into the local optimizer."""
-        name_list = [n for u in self.units for n in u.param_names]
+        name_to_idx = {}
+        for u in self.units:
+            for n in u.param_names:
+                name_to_idx[n] = len(name_to_idx)
         for group in optimizer.param_groups:
             for shard_p in group["params"]:
                 u = next((x for x in self.units if x.shard is shard_p), None)
@@ -464,7 +595,7 @@ class StokeFSDPModule(nn.Module):
                 # re-flatten; scalar tensors (e.g. Adam's `step`) pass through.
                 keys = set()
                 for p, name in zip(u.params, u.param_names):
-                    st = full["state"].get(name_list.index(name))
+                    st = full["state"].get(name_to_idx[name])
                     if st:
                         keys.update(
                             k for k, v in st.items()
@@ -477,7 +608,7 @@ class StokeFSDPModule(nn.Module):
                         u.padded, dtype=torch.float32, device=u.shard.device
                     )
                     for p, name, o in zip(u.params, u.param_names, u.offsets):
-                        st = full["state"].get(name_list.index(name))
+                        st = full["state"].get(name_to_idx[name])
                         if st and k in st:
                             flat[o : o + p.numel()].copy_(
                                 st[k].reshape(-1).float().to(u.shard.device)
@@ -488,7 +619,7 @@ class StokeFSDPModule(nn.Module):
                     ].clone()
                     del flat
                 # Scalar entries (e.g. step) come from the first param
-                st0 = full["state"].get(name_list.index(u.param_names[0]), {})
+                st0 = full["state"].get(name_to_idx[u.param_names[0]], {})
                 for k, v in st0.items():
                     if k in state_entry:
                         continue
