@@ -1,26 +1,26 @@
-// Flash-attention forward (FA-2 style) for MI355X — ROUND-2 WORK IN
-// PROGRESS, correctness-first v0.
+// Flash-attention forward (FA-2 style) for MI355X gfx950 — v1.
 //
-// Status: compiles for gfx950 and is exercised ONLY by the env-gated tests
-// in tests/test_fa_wip.py (STOKE_FA_TEST=1); nothing in the framework or
-// benchmarks calls it yet.  The production path remains
-// F.scaled_dot_product_attention (AOTriton).  See NOTES.md for the design
-// and the round-2 plan (backward, LDS double-buffering, larger tiles).
+// v1 follows the measured CDNA4 attention structure from the platform
+// guide's ladder (8-wave 32x32 MFMA, swapped QK^T so softmax is lane-local,
+// XOR-swizzled K image, transposed V image, in-register P->bf16):
+//   * block = 8 waves x 32 q rows = 256-row Q tile; KV walked in 64-row
+//     tiles staged in LDS once per block (shared by all 8 waves).
+//   * S^T = K Q^T via v_mfma_f32_32x32x16_bf16: the C fragment then holds,
+//     per lane, one q COLUMN and 16 kv rows -> the softmax row-reduce is
+//     15 in-lane fmax + one lane<->lane+32 exchange, no LDS round trip.
+//   * K tile LDS image is row-major with a 16-B-slot XOR swizzle
+//     (slot ^= row&15) so the A-fragment ds_read_b128 is bank-conflict-free.
+//   * V is staged TRANSPOSED ([d][kv], padded) so the PV A-fragment
+//     (V^T[d][kv]) is a contiguous b128 read; O accumulates as O^T[d][q].
+//   * P (f32, C layout) is repacked to bf16 A/B fragments in registers:
+//     v_cvt_pk_bf16_f32 pairs + one __shfl_xor(32) half-exchange per pair
+//     of dwords — no LDS staging of P.
 //
-// v0 design (one wave = 16 query rows; 4 independent waves per block):
-//   * S = Q K^T via __builtin_amdgcn_mfma_f32_16x16x32_bf16 over head-dim
-//     chunks of 32; KV walked in 32-column tiles (two 16x16 S quadrants).
-//   * online softmax: per-lane running rowmax/rowsum for the wave's 4
-//     C-rows, cross-lane reduced over the 16 lanes of each row group.
-//   * P staged through LDS to convert the C-fragment layout into the
-//     A-fragment layout for the P x V MFMAs.
-//   * causal masking per element; GQA by head-index mapping.
-//
-// MFMA fragment-layout assumptions (validated by the mfma_probe_* tests
-// before anything else — the guide defers exact A/B maps to the ISA doc):
-//   A[16x32]: lane l holds row (l & 15), k = (l >> 4)*8 + e,  e = 0..7
-//   B[32x16]: lane l holds col (l & 15), k = (l >> 4)*8 + e
-//   C[16x16]: lane l holds col (l & 15), row (l >> 4)*4 + r,  r = 0..3
+// Fragment maps (validated on hardware by tests/test_fa_wip.py probes):
+//   16x16x32: A row l&15, k=(l>>4)*8+e; B col l&15 same k;
+//             C col l&15, row (l>>4)*4+r.
+//   32x32x16: A row l&31, k=(l>>5)*8+e; B col l&31 same k;
+//             C col l&31, row (r&3)+8*(r>>2)+4*(l>>5), r=0..15.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -31,29 +31,32 @@
 namespace {
 
 typedef __hip_bfloat16 bf16;
-typedef __attribute__((ext_vector_type(8))) short bf16x8v;  // MFMA A/B frag
-typedef __attribute__((ext_vector_type(4))) float f32x4v;   // MFMA C/D frag
-
-__device__ __forceinline__ float fa_b2f(short v) {
-  __hip_bfloat16_raw r;
-  r.x = (unsigned short)v;
-  return __bfloat162float(*reinterpret_cast<bf16*>(&r));
-}
-
-__device__ __forceinline__ short fa_f2b(float f) {
-  bf16 h = __float2bfloat16(f);
-  return (short)*reinterpret_cast<unsigned short*>(&h);
-}
+typedef __attribute__((ext_vector_type(8))) short bf16x8v;   // MFMA A/B frag
+typedef __attribute__((ext_vector_type(4))) float f32x4v;    // 16x16 C frag
+typedef __attribute__((ext_vector_type(16))) float f32x16v;  // 32x32 C frag
 
 __device__ __forceinline__ bf16 fa_f2bf(float f) {
   return __float2bfloat16(f);
 }
 
+__device__ __forceinline__ float fa_bf2f(short v) {
+  __hip_bfloat16_raw r;
+  r.x = (unsigned short)v;
+  return __bfloat162float(*reinterpret_cast<bf16*>(&r));
+}
+
+// pack two f32 -> two bf16 in one dword (round-to-nearest-even)
+__device__ __forceinline__ unsigned int pk_bf16(float lo, float hi) {
+  bf16 a = __float2bfloat16(lo);
+  bf16 b = __float2bfloat16(hi);
+  unsigned int r = (unsigned int)*reinterpret_cast<unsigned short*>(&a);
+  r |= ((unsigned int)*reinterpret_cast<unsigned short*>(&b)) << 16;
+  return r;
+}
+
 // ---------------------------------------------------------------------------
-// Layout probe: D = A x B for one 16x16x32 MFMA with fragments loaded per the
-// assumed lane maps.  The host test checks against a torch fp32 matmul with
-// ASYMMETRIC operands, so a wrong map fails loudly (and tells round 2 what
-// to fix before any attention debugging).
+// Layout probes (host tests check against torch matmuls with ASYMMETRIC
+// operands, per the guide: symmetric inputs hide row/col swaps).
 // ---------------------------------------------------------------------------
 __global__ void mfma_probe_16x16x32(const bf16* __restrict__ A,  // [16,32] rm
                                     const bf16* __restrict__ B,  // [32,16] rm
@@ -73,155 +76,227 @@ __global__ void mfma_probe_16x16x32(const bf16* __restrict__ A,  // [16,32] rm
     D[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
 }
 
+__global__ void mfma_probe_32x32x16(const bf16* __restrict__ A,  // [32,16] rm
+                                    const bf16* __restrict__ B,  // [16,32] rm
+                                    float* __restrict__ D) {     // [32,32] rm
+  const int l = threadIdx.x;  // 64 lanes
+  bf16x8v a, b;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int k = (l >> 5) * 8 + e;
+    a[e] = (short)*reinterpret_cast<const unsigned short*>(&A[(l & 31) * 16 + k]);
+    b[e] = (short)*reinterpret_cast<const unsigned short*>(&B[k * 32 + (l & 31)]);
+  }
+  f32x16v c;
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) c[r] = 0.f;
+  c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+    D[row * 32 + (l & 31)] = c[r];
+  }
+}
+
 // ---------------------------------------------------------------------------
-// FA forward v0.  q: [B,H,S,Dh], k/v: [B,Hkv,S,Dh] bf16 contiguous,
+// FA forward v1.  q: [B,H,S,Dh], k/v: [B,Hkv,S,Dh] bf16 contiguous,
 // Dh in {64, 128}; out: [B,H,S,Dh] bf16; lse: [B,H,S] fp32.
-// Grid: (ceil(S/64), B*H); block 256 = 4 waves, wave w owns q rows
-// [blk*64 + w*16, +16).
+// Grid: (ceil(S/256), B*H); block 512 = 8 waves, wave w owns q rows
+// [blk*256 + w*32, +32).
 // ---------------------------------------------------------------------------
+#define FA_KVB 64  // kv rows per staged tile
+#define FA_VPAD 8  // V^T row padding (elements) to spread write banks
+
 template <int DH>
-__global__ __launch_bounds__(256) void fa_fwd_kernel(
+__global__ __launch_bounds__(512) void fa_fwd_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     float* __restrict__ lse, int S, int H, int HKV, int causal) {
-  constexpr int DC = DH / 32;  // head-dim chunks per MFMA K
+  constexpr int DCH = DH / 16;    // 16-wide k-chunks of the head dim
+  constexpr int DT = DH / 32;     // 32-row d-tiles of O^T
+  constexpr int KSLOT = DH / 8;   // 16-B slots per K row
   const int wave = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int row0 = blockIdx.x * 64 + wave * 16;  // first q row of this wave
+  const int lq = l & 31;          // this lane's q column / d row
+  const int h2 = l >> 5;          // lane half (0: lanes 0-31, 1: 32-63)
   const int bh = blockIdx.y;
   const int h = bh % H;
   const int hkv = h / (H / HKV);
   const int b = bh / H;
-  // NOTE: no early return — every wave must reach the block barriers below;
-  // out-of-range rows are clamped and their stores guarded.
-  const bool wave_active = row0 < S;
+  const int qbase = blockIdx.x * 256;
+  const int row0 = qbase + wave * 32;          // wave's first q row
+  const int qrow = min(row0 + lq, S - 1);      // this lane's q row (clamped)
+
   const bf16* qp = q + (((long)b * H + h) * S) * DH;
   const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
   const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
 
-  // Q fragments for this wave's 16 rows, kept in registers for the whole
-  // KV sweep.  Lane l -> row (l&15); rows past S replicate the last valid
-  // row (their outputs are never stored).
-  const int qrow = min(min(row0 + (l & 15), S - 1), S - 1);
-  bf16x8v qfrag[DC];
-  #pragma unroll
-  for (int c = 0; c < DC; ++c) {
-    const bf16* src = qp + (long)qrow * DH + c * 32 + (l >> 4) * 8;
-    qfrag[c] = *reinterpret_cast<const bf16x8v*>(src);
-  }
+  // K image: row-major [KVB][DH], 16-B slot index XOR-swizzled with row&15.
+  // V image: transposed [DH][KVB + pad].
+  __shared__ bf16 kbuf[FA_KVB * DH];
+  __shared__ bf16 vbuf[DH * (FA_KVB + FA_VPAD)];
 
-  float m[4], lsum[4];
+  // Q as B-operand fragments, resident for the whole KV sweep:
+  // frag ch: lane holds q col lq, k-elems d = ch*16 + h2*8 + e.
+  bf16x8v qf[DCH];
   #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m[r] = -1e30f;
-    lsum[r] = 0.f;
-  }
-  f32x4v oacc[DH / 16];
+  for (int ch = 0; ch < DCH; ++ch)
+    qf[ch] = *reinterpret_cast<const bf16x8v*>(
+        qp + (long)qrow * DH + ch * 16 + h2 * 8);
+
+  f32x16v oacc[DT];
   #pragma unroll
-  for (int f = 0; f < DH / 16; ++f) oacc[f] = {0.f, 0.f, 0.f, 0.f};
+  for (int dt = 0; dt < DT; ++dt)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[dt][r] = 0.f;
+  float m = -1e30f, lsum = 0.f;
+  const float scale = rsqrtf((float)DH);
 
-  // LDS: per-wave P tile [16 rows][32 kcols] bf16
-  __shared__ bf16 p_lds[4][16][32];
-
-  // All waves in the block iterate the same KV range (to the block's last
-  // row under causal masking) so __syncthreads stays uniform.
-  const int block_last_row = min(blockIdx.x * 64 + 63, S - 1);
-  const int kv_end = causal ? (block_last_row + 1) : S;
-  for (int kb = 0; kb < kv_end; kb += 32) {
-    // ---- S quadrants: Sq[16][16] for kcol halves 0 and 1
-    float srows[2][4];  // [quadrant][C row] for this lane
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd) {
-      f32x4v acc = {0.f, 0.f, 0.f, 0.f};
-      const int kcol = min(kb + qd * 16 + (l & 15), S - 1);
-      #pragma unroll
-      for (int c = 0; c < DC; ++c) {
-        bf16x8v bfrag;
-        const bf16* src = kp + (long)kcol * DH + c * 32 + (l >> 4) * 8;
-        bfrag = *reinterpret_cast<const bf16x8v*>(src);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfrag, acc,
-                                                      0, 0, 0);
+  const int kv_end = causal ? min(qbase + 256, S) : S;
+  for (int kb = 0; kb < kv_end; kb += FA_KVB) {
+    // ---- stage K (swizzled rows) and V^T cooperatively: 512 threads x
+    // 8-elem pieces.  K tile: 64*DH/8 pieces; V tile the same count.
+    __syncthreads();
+    {
+      const int npiece = FA_KVB * DH / 8;
+      for (int p = threadIdx.x; p < npiece; p += 512) {
+        const int row = p / KSLOT;          // kv row
+        const int slot = p % KSLOT;         // 16-B slot within row
+        const int grow = min(kb + row, S - 1);
+        bf16x8v piece = *reinterpret_cast<const bf16x8v*>(
+            kp + (long)grow * DH + slot * 8);
+        const int sslot = slot ^ (row & 15);
+        *reinterpret_cast<bf16x8v*>(&kbuf[row * DH + sslot * 8]) = piece;
+        // V: same global shape; scatter-transpose into vbuf[d][kv]
+        union { bf16x8v v8; short s[8]; } vpiece;
+        vpiece.v8 = *reinterpret_cast<const bf16x8v*>(
+            vp + (long)grow * DH + slot * 8);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vbuf[(slot * 8 + j) * (FA_KVB + FA_VPAD) + row] =
+              *reinterpret_cast<const bf16*>(&vpiece.s[j]);
       }
-      const float scale = rsqrtf((float)DH);
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int rr = row0 + (l >> 4) * 4 + r;      // this C-elem's q row
-        const int cc = kb + qd * 16 + (l & 15);      // its k column
-        float s = acc[r] * scale;
-        if (cc >= S || rr >= S || (causal && cc > rr)) s = -1e30f;
-        srows[qd][r] = s;
-      }
-    }
-    // ---- online softmax: row max over the 16 lanes of each row group
-    float mnew[4];
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mx = fmaxf(srows[0][r], srows[1][r]);
-      #pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, 16));
-      mnew[r] = fmaxf(m[r], mx);
-    }
-    float alpha[4];
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      alpha[r] = __expf(m[r] - mnew[r]);
-      m[r] = mnew[r];
-    }
-    // ---- P = exp(S - m); row sums; stage P into LDS in A-layout order
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd) {
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float p = __expf(srows[qd][r] - m[r]);
-        srows[qd][r] = p;
-        p_lds[wave][(l >> 4) * 4 + r][qd * 16 + (l & 15)] = fa_f2bf(p);
-      }
-    }
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float ps = srows[0][r] + srows[1][r];
-      #pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        ps += __shfl_xor(ps, off, 16);
-      lsum[r] = lsum[r] * alpha[r] + ps;
     }
     __syncthreads();
-    // ---- O = O*alpha + P x V  (P from LDS in A layout; V B-frags global)
-    bf16x8v pfrag;
+
+    // ---- S^T tiles: sacc[ct] = K[ct*32..+32] x Q^T  (C: lane=q col,
+    // 16 kv rows each); mask + scale in-register.
+    float p32[32];  // this lane's P values, kv = ct*32 + (r&3)+8*(r>>2)+4*h2
+    float tmax = -1e30f;
     #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      pfrag[e] = (short)*reinterpret_cast<unsigned short*>(
-          &p_lds[wave][l & 15][(l >> 4) * 8 + e]);
-    #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
+    for (int ct = 0; ct < 2; ++ct) {
+      f32x16v sacc;
       #pragma unroll
-      for (int r = 0; r < 4; ++r) oacc[f][r] *= alpha[r];
-      bf16x8v vfrag;
+      for (int r = 0; r < 16; ++r) sacc[r] = 0.f;
       #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int krow = min(kb + (l >> 4) * 8 + e, S - 1);
-        vfrag[e] = (short)*reinterpret_cast<const unsigned short*>(
-            &vp[(long)krow * DH + f * 16 + (l & 15)]);
+      for (int ch = 0; ch < DCH; ++ch) {
+        // A-frag: K row kv = ct*32 + lq, k-elems d = ch*16 + h2*8 + e
+        const int krow = ct * 32 + lq;
+        const int slot = (ch * 2 + h2) ^ (krow & 15);
+        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+            &kbuf[krow * DH + slot * 8]);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ch], sacc,
+                                                       0, 0, 0);
       }
-      oacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, oacc[f],
-                                                        0, 0, 0);
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvg = kb + ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+        const int qg = row0 + lq;
+        float s = sacc[r] * scale;
+        if (kvg >= S || qg >= S || (causal && kvg > qg)) s = -1e30f;
+        p32[ct * 16 + r] = s;
+        tmax = fmaxf(tmax, s);
+      }
     }
-    __syncthreads();
-  }
-  // ---- epilogue: O /= lsum; store O (bf16) and logsumexp (fp32)
-  #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int rr = row0 + (l >> 4) * 4 + r;
-    if (!wave_active || rr >= S) continue;
-    const float inv = 1.f / fmaxf(lsum[r], 1e-30f);
+    // ---- online softmax (lane-local rows; combine lane<->lane+32)
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
+    const float mnew = fmaxf(m, tmax);
+    const float alpha = __expf(m - mnew);
+    m = mnew;
+    float psum = 0.f;
     #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
-      out[(((long)b * H + h) * S + rr) * DH + f * 16 + (l & 15)] =
-          fa_f2bf(oacc[f][r] * inv);
+    for (int i = 0; i < 32; ++i) {
+      const float p = __expf(p32[i] - mnew);
+      p32[i] = p;
+      psum += p;
     }
-    if ((l & 15) == 0)
-      lse[((long)b * H + h) * S + rr] = m[r] + __logf(fmaxf(lsum[r], 1e-30f));
+    psum += __shfl_xor(psum, 32);
+    lsum = lsum * alpha + psum;
+    #pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[dt][r] *= alpha;
+
+    // ---- P -> bf16 B-fragments in registers.  Fragment kch: lane needs
+    // kv = kch*16 + h2*8 + e; elements e<4 live in the h=0 lane of this q,
+    // e>=4 in the h=1 lane.  Each lane packs the 4 values it owns for both
+    // destination halves, then one half-exchange per dword pair.
+    bf16x8v pa[FA_KVB / 16];
+    #pragma unroll
+    for (int kch = 0; kch < FA_KVB / 16; ++kch) {
+      unsigned int d01[2], d23[2];  // prep for dest half 0 / half 1
+      #pragma unroll
+      for (int ht = 0; ht < 2; ++ht) {
+        float pv[4];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          // kv = kch*16 + ht*8 + h2*4 + j; h2 is lane-dependent, so index
+          // p32 with BOTH compile-time candidates and cndmask-select —
+          // a runtime index would spill p32 to scratch.
+          const int kv0 = kch * 16 + ht * 8 + j;       // h2 == 0
+          const int kv1 = kv0 + 4;                     // h2 == 1
+          const int i0 = (kv0 >> 5) * 16 + ((kv0 & 31) & 3) + 4 * ((kv0 & 31) >> 3);
+          const int i1 = (kv1 >> 5) * 16 + ((kv1 & 31) & 3) + 4 * ((kv1 & 31) >> 3);
+          pv[j] = h2 ? p32[i1] : p32[i0];
+        }
+        unsigned int* dst = ht == 0 ? d01 : d23;
+        dst[0] = pk_bf16(pv[0], pv[1]);
+        dst[1] = pk_bf16(pv[2], pv[3]);
+      }
+      const unsigned int s01_0 = __shfl_xor((int)d01[0], 32);
+      const unsigned int s01_1 = __shfl_xor((int)d01[1], 32);
+      const unsigned int s23_0 = __shfl_xor((int)d23[0], 32);
+      const unsigned int s23_1 = __shfl_xor((int)d23[1], 32);
+      unsigned int w0, w1, w2, w3;
+      if (h2 == 0) {
+        w0 = d01[0]; w1 = d01[1]; w2 = s01_0; w3 = s01_1;
+      } else {
+        w0 = s23_0; w1 = s23_1; w2 = d23[0]; w3 = d23[1];
+      }
+      unsigned int frag[4] = {w0, w1, w2, w3};
+      pa[kch] = *reinterpret_cast<bf16x8v*>(frag);
+    }
+
+    // ---- O^T += V^T x P^T   (A: V^T[d][kv] contiguous b128 from vbuf)
+    #pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      #pragma unroll
+      for (int kch = 0; kch < FA_KVB / 16; ++kch) {
+        bf16x8v vf = *reinterpret_cast<const bf16x8v*>(
+            &vbuf[(dt * 32 + lq) * (FA_KVB + FA_VPAD) + kch * 16 + h2 * 8]);
+        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pa[kch],
+                                                           oacc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: O^T[d][q] /= lsum; store out[q][d] (paired 4-B stores),
+  // lse[q] = m + log(lsum).
+  const int qg = row0 + lq;
+  if (qg < S) {
+    const float inv = 1.f / fmaxf(lsum, 1e-30f);
+    bf16* op = out + (((long)b * H + h) * S + qg) * DH;
+    #pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      #pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+        *reinterpret_cast<unsigned int*>(op + d) =
+            pk_bf16(oacc[dt][r] * inv, oacc[dt][r + 1] * inv);
+      }
+    }
+    if (h2 == 0)
+      lse[((long)b * H + h) * S + qg] = m + __logf(fmaxf(lsum, 1e-30f));
   }
 }
 
@@ -234,6 +309,13 @@ void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {
                      D.data_ptr<float>());
 }
 
+void mfma_probe32(at::Tensor A, at::Tensor B, at::Tensor D) {
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(mfma_probe_32x32x16, dim3(1), dim3(64), 0, stream,
+                     (const bf16*)A.data_ptr(), (const bf16*)B.data_ptr(),
+                     D.data_ptr<float>());
+}
+
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                bool causal) {
   TORCH_CHECK(q.scalar_type() == at::kBFloat16 && q.dim() == 4 &&
@@ -241,19 +323,19 @@ std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
               "fa_fwd: contiguous [B,H,S,D] bf16");
   const int B = q.size(0), H = q.size(1), S = q.size(2), DH = q.size(3);
   const int HKV = k.size(1);
-  TORCH_CHECK(DH == 64 || DH == 128, "fa_fwd v0: head dim 64 or 128");
+  TORCH_CHECK(DH == 64 || DH == 128, "fa_fwd: head dim 64 or 128");
   TORCH_CHECK(H % HKV == 0, "fa_fwd: H must be a multiple of H_kv");
   auto out = at::empty_like(q);
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
-  dim3 grid((S + 63) / 64, B * H);
+  dim3 grid((S + 255) / 256, B * H);
   if (DH == 128) {
-    hipLaunchKernelGGL((fa_fwd_kernel<128>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_fwd_v1<128>), grid, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
                        lse.data_ptr<float>(), S, H, HKV, causal ? 1 : 0);
   } else {
-    hipLaunchKernelGGL((fa_fwd_kernel<64>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_fwd_v1<64>), grid, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
                        lse.data_ptr<float>(), S, H, HKV, causal ? 1 : 0);

@@ -669,8 +669,9 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor x, at::Tensor dy,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor invstd);
 
-// Flash-attention forward — ROUND-2 WIP, env-gated tests only (csrc/fa_fwd.hip)
+// Flash-attention forward (csrc/fa_fwd.hip)
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D);
+void mfma_probe32(at::Tensor A, at::Tensor B, at::Tensor D);
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                bool causal);
 std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
@@ -698,6 +699,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd, "fused bf16 SwiGLU backward");
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
+  m.def("mfma_probe32", &mfma_probe32,
+        "layout probe: one v_mfma_f32_32x32x16_bf16 (A[32,16] x B[16,32])");
   m.def("mfma_probe", &mfma_probe,
         "16x16x32 bf16 MFMA fragment-layout probe (round-2 WIP)");
   m.def("fa_fwd", &fa_fwd,

@@ -50,11 +50,33 @@ def test_mfma_probe_layout():
     )
 
 
+def test_mfma_probe32_layout():
+    """Validates the 32x32x16 A/B/C lane maps the v1 kernel assumes."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.manual_seed(0)
+    A = (torch.arange(32 * 16, device="cuda").float().reshape(32, 16)
+         % 7 - 3).bfloat16() * 0.25
+    B = (torch.arange(16 * 32, device="cuda").float().reshape(16, 32)
+         % 5 - 2).bfloat16() * 0.5
+    D = torch.zeros(32, 32, device="cuda", dtype=torch.float32)
+    _ext().mfma_probe32(A, B, D)
+    torch.cuda.synchronize()
+    want = A.float() @ B.float()
+    err = (D - want).abs().max().item()
+    assert err < 1e-2, (
+        f"32x32x16 fragment-layout mismatch (max err {err}): fix the lane "
+        "maps in csrc/fa_fwd.hip before debugging attention"
+    )
+
+
 @pytest.mark.parametrize("B,H,HKV,S,D,causal", [
     (1, 2, 2, 64, 64, False),
     (1, 2, 2, 128, 128, True),
     (2, 4, 2, 96, 128, True),   # GQA + ragged S
     (1, 1, 1, 300, 64, True),
+    (1, 2, 1, 1024, 64, True),  # multi-256-block path (v1 grid)
+    (1, 2, 2, 500, 128, True),  # ragged across block boundary
 ])
 def test_fa_fwd_vs_sdpa(B, H, HKV, S, D, causal):
     if not torch.cuda.is_available():
