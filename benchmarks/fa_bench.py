@@ -65,8 +65,8 @@ def run(tag, B, H, HKV, S, D):
     t_fa_fb = bench(fa_fb)
     # rough flops: fwd 4*B*H*S^2*D/2 (causal), bwd ~2.5x fwd
     fl = 4 * B * H * S * S * D / 2
-    print(f"{tag}: fwd sdpa {t_sdpa_f:.3f} ms ({fl/t_sdpa_f/1e9:.0f} GF/s) "
-          f"| fa {t_fa_f:.3f} ms ({fl/t_fa_f/1e9:.0f} GF/s)")
+    print(f"{tag}: fwd sdpa {t_sdpa_f:.3f} ms ({fl/t_sdpa_f/1e9:.0f} TF/s) "
+          f"| fa {t_fa_f:.3f} ms ({fl/t_fa_f/1e9:.0f} TF/s)")
     print(f"{tag}: f+b sdpa {t_sdpa_fb:.3f} ms | fa {t_fa_fb:.3f} ms "
           f"| speedup {t_sdpa_fb/t_fa_fb:.2f}x")
 

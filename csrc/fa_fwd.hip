@@ -152,32 +152,74 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   float m = -1e30f, lsum = 0.f;
   const float scale = rsqrtf((float)DH);
 
-  const int kv_end = causal ? min(qbase + 256, S) : S;
-  for (int kb = 0; kb < kv_end; kb += FA_KVB) {
-    // ---- stage K (swizzled rows) and V^T cooperatively: 512 threads x
-    // 8-elem pieces.  K tile: 64*DH/8 pieces; V tile the same count.
-    __syncthreads();
-    {
-      const int npiece = FA_KVB * DH / 8;
-      for (int p = threadIdx.x; p < npiece; p += 512) {
-        const int row = p / KSLOT;          // kv row
-        const int slot = p % KSLOT;         // 16-B slot within row
+  // ---- tile staging (T14 register split: global loads for tile t+1 are
+  // issued during tile t's compute; the LDS writes happen after the
+  // end-of-tile barrier).  Thread halves specialize: tid<256 stages K
+  // (b128 loads -> swizzled b128 LDS writes), tid>=256 stages V
+  // (4 kv-rows of one 8-wide d-chunk -> 8 ds_write_b64 into the
+  // transposed image; 16 lanes share a d-row so write banks are distinct).
+  constexpr int KPIECE = FA_KVB * KSLOT / 256;  // K b128 pieces per thread
+  const int tid = threadIdx.x;
+  const bool is_k = tid < 256;
+  const int ka = tid & 255;
+  const int v_dchunk = ka >> 4;       // V: which 8-wide d chunk
+  const int v_kvq = ka & 15;          // V: which 4-row kv quad
+  const bool v_active = !is_k && v_dchunk < DH / 8;
+  bf16x8v stage[4];                   // K: KPIECE pieces; V: 4 kv rows
+
+  auto load_tile = [&](int kb) {
+    if (is_k) {
+      #pragma unroll
+      for (int i = 0; i < KPIECE; ++i) {
+        const int p = ka * KPIECE + i;
+        const int row = p / KSLOT;
+        const int slot = p % KSLOT;
         const int grow = min(kb + row, S - 1);
-        bf16x8v piece = *reinterpret_cast<const bf16x8v*>(
+        stage[i] = *reinterpret_cast<const bf16x8v*>(
             kp + (long)grow * DH + slot * 8);
-        const int sslot = slot ^ (row & 15);
-        *reinterpret_cast<bf16x8v*>(&kbuf[row * DH + sslot * 8]) = piece;
-        // V: same global shape; scatter-transpose into vbuf[d][kv]
-        union { bf16x8v v8; short s[8]; } vpiece;
-        vpiece.v8 = *reinterpret_cast<const bf16x8v*>(
-            vp + (long)grow * DH + slot * 8);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vbuf[(slot * 8 + j) * (FA_KVB + FA_VPAD) + row] =
-              *reinterpret_cast<const bf16*>(&vpiece.s[j]);
+      }
+    } else if (v_active) {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int grow = min(kb + v_kvq * 4 + i, S - 1);
+        stage[i] = *reinterpret_cast<const bf16x8v*>(
+            vp + (long)grow * DH + v_dchunk * 8);
       }
     }
-    __syncthreads();
+  };
+
+  auto write_tile = [&]() {
+    if (is_k) {
+      #pragma unroll
+      for (int i = 0; i < KPIECE; ++i) {
+        const int p = ka * KPIECE + i;
+        const int row = p / KSLOT;
+        const int slot = p % KSLOT;
+        const int sslot = slot ^ (row & (KSLOT - 1));
+        *reinterpret_cast<bf16x8v*>(&kbuf[row * DH + sslot * 8]) = stage[i];
+      }
+    } else if (v_active) {
+      // stage[i] holds V[kv0+i][d0..d0+7]; emit b64 of 4 kv values per d
+      union { bf16x8v v8[4]; short s[4][8]; } u;
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) u.v8[i] = stage[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { short s4[4]; unsigned long long d; } pack;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) pack.s4[i] = u.s[i][j];
+        *reinterpret_cast<unsigned long long*>(
+            &vbuf[(v_dchunk * 8 + j) * (FA_KVB + FA_VPAD) + v_kvq * 4]) =
+            pack.d;
+      }
+    }
+  };
+
+  const int kv_end = causal ? min(qbase + 256, S) : S;
+  load_tile(0);
+  write_tile();
+  __syncthreads();
+  for (int kb = 0; kb < kv_end; kb += FA_KVB) {
 
     // ---- S^T tiles: sacc[ct] = K[ct*32..+32] x Q^T  (C: lane=q col,
     // 16 kv rows each); mask + scale in-register.
@@ -192,7 +234,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
       for (int ch = 0; ch < DCH; ++ch) {
         // A-frag: K row kv = ct*32 + lq, k-elems d = ch*16 + h2*8 + e
         const int krow = ct * 32 + lq;
-        const int slot = (ch * 2 + h2) ^ (krow & 15);
+        const int slot = (ch * 2 + h2) ^ (krow & (KSLOT - 1));
         bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
             &kbuf[krow * DH + slot * 8]);
         sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ch], sacc,
@@ -208,6 +250,10 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
         tmax = fmaxf(tmax, s);
       }
     }
+    // issue next tile's global loads now — they complete under the
+    // softmax + PV compute below (T14 async-stage split)
+    if (kb + FA_KVB < kv_end) load_tile(kb + FA_KVB);
+
     // ---- online softmax (lane-local rows; combine lane<->lane+32)
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
     const float mnew = fmaxf(m, tmax);
@@ -277,6 +323,11 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
         oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pa[kch],
                                                            oacc[dt], 0, 0, 0);
       }
+    }
+    __syncthreads();  // every wave done reading this tile's K/V images
+    if (kb + FA_KVB < kv_end) {
+      write_tile();
+      __syncthreads();
     }
   }
 
