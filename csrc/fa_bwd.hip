@@ -1,18 +1,28 @@
-// Flash-attention backward (FA-2 split) for MI355X — ROUND-2 WORK IN
-// PROGRESS, correctness-first v0.  Same status and gating as fa_fwd.hip:
-// compiles for gfx950, exercised only by STOKE_FA_TEST=1 tests, nothing in
-// the framework calls it.  Uses the identical MFMA fragment-layout
-// assumptions (validated first by the mfma_probe test).
+// Flash-attention backward (FA-2 split) for MI355X gfx950 — v1.
 //
-// Standard FA-2 decomposition with saved logsumexp L and
-// delta = rowsum(dO * O):
-//   P    = exp(S*scale - L)
-//   dV  += P^T dO
-//   dP   = dO V^T;   dS = P * (dP - delta) * scale
-//   dK  += dS^T Q;   dQ += dS K
-// Two kernels: fa_bwd_dkv (grid over KV 16-row waves, streams Q/dO tiles)
-// and fa_bwd_dq (grid over Q 16-row waves, streams K/V tiles); plus a tiny
-// delta kernel.  No atomics: each wave owns its dK/dV (resp. dQ) rows.
+// Same 8-wave 32x32-MFMA structure as the v1 forward (csrc/fa_fwd.hip):
+// LDS-staged tiles shared by all 8 waves, XOR-swizzled row-major images for
+// row-operand (A) reads, padded transposed images for the d-major (A^T)
+// reads, and the C->B fragment half-exchange repack (cvt_pk + shfl_xor 32)
+// so P / dS never round-trip through LDS.
+//
+// Math (standard FA-2 with saved logsumexp L and delta = rowsum(dO*O)):
+//   P  = exp(S*scale - L)
+//   dV += P^T dO        dP = dO V^T
+//   dS = P * (dP - delta) * scale
+//   dK += dS^T Q        dQ += dS K
+//
+// Kernel split:
+//   fa_bwd_dkv_v1: block owns 256 kv rows (8 waves x 32); loops over the
+//     q-heads of its kv head (GQA accumulated IN REGISTERS — no fp32
+//     scratch buffers, no host-side reduction) and over 64-row q tiles.
+//     Per tile it stages Q and dO row-major (swizzled) AND transposed
+//     (padded), computes C[q,kv] products (lane = kv column, so the wave's
+//     own K/V registers serve as the B operand), and accumulates
+//     dV^T/dK^T[d,kv] with transposed-image A fragments.
+//   fa_bwd_dq_v1: block owns 256 q rows; mirrors the forward (K/V staged,
+//     Q/dO in registers, C[kv,q] products with lane = q column, dQ^T
+//     accumulated from the transposed K image).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -24,7 +34,10 @@ namespace {
 
 typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) short bf16x8v;
-typedef __attribute__((ext_vector_type(4))) float f32x4v;
+typedef __attribute__((ext_vector_type(16))) float f32x16v;
+
+#define FB_QB 64    // q rows staged per dkv tile / kv rows per dq tile
+#define FB_PAD 8    // transposed-image row padding (elements)
 
 __device__ __forceinline__ bf16 fb_f2bf(float f) {
   return __float2bfloat16(f);
@@ -34,280 +47,416 @@ __device__ __forceinline__ float fb_b2f(const bf16& h) {
   return __bfloat162float(h);
 }
 
+__device__ __forceinline__ unsigned int fb_pk(float lo, float hi) {
+  bf16 a = __float2bfloat16(lo);
+  bf16 b = __float2bfloat16(hi);
+  unsigned int r = (unsigned int)*reinterpret_cast<unsigned short*>(&a);
+  r |= ((unsigned int)*reinterpret_cast<unsigned short*>(&b)) << 16;
+  return r;
+}
+
+// Repack one 32-wide C tile (16 f32 regs, rows = reduction dim) into two
+// bf16 B/A fragments whose k-elems run along that dim.  Identical to the
+// forward's repack: prep both destination halves, one half-exchange.
+__device__ __forceinline__ void fb_repack16(const f32x16v& pr, int h2,
+                                            bf16x8v out[2]) {
+  #pragma unroll
+  for (int kc = 0; kc < 2; ++kc) {
+    unsigned int d01[2], d23[2];
+    #pragma unroll
+    for (int ht = 0; ht < 2; ++ht) {
+      float pv[4];
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        // reduction index = kc*16 + ht*8 + h2*4 + j; reg = j + 4*(kc*2+ht)
+        // is independent of h2 only through the value selection below
+        const int r0 = j + 4 * (kc * 2 + ht);
+        pv[j] = pr[r0];
+      }
+      unsigned int* dst = ht == 0 ? d01 : d23;
+      dst[0] = fb_pk(pv[0], pv[1]);
+      dst[1] = fb_pk(pv[2], pv[3]);
+    }
+    const unsigned int s01_0 = __shfl_xor((int)d01[0], 32);
+    const unsigned int s01_1 = __shfl_xor((int)d01[1], 32);
+    const unsigned int s23_0 = __shfl_xor((int)d23[0], 32);
+    const unsigned int s23_1 = __shfl_xor((int)d23[1], 32);
+    unsigned int w[4];
+    if (h2 == 0) {
+      w[0] = d01[0]; w[1] = d01[1]; w[2] = s01_0; w[3] = s01_1;
+    } else {
+      w[0] = s23_0; w[1] = s23_1; w[2] = d23[0]; w[3] = d23[1];
+    }
+    out[kc] = *reinterpret_cast<bf16x8v*>(w);
+  }
+}
+
 // delta[b,h,s] = sum_d dO[b,h,s,d] * O[b,h,s,d]  (fp32)
 __global__ __launch_bounds__(256) void fa_delta_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ out,
     float* __restrict__ delta, long rows, int DH) {
-  // one wave per row, lanes stride the head dim
   const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
   const int l = threadIdx.x & 63;
   if (row >= rows) return;
   float s = 0.f;
-  for (int d = l; d < DH; d += 64)
+  for (int d = l * 2; d < DH; d += 128) {
     s += fb_b2f(dout[row * DH + d]) * fb_b2f(out[row * DH + d]);
+    s += fb_b2f(dout[row * DH + d + 1]) * fb_b2f(out[row * DH + d + 1]);
+  }
   #pragma unroll
   for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off);
   if (l == 0) delta[row] = s;
 }
 
 // ---------------------------------------------------------------------------
-// dK/dV kernel: wave owns 16 KV rows; iterates q tiles of 32 (from the
-// causal diagonal onward).  Works in the transposed frame: C rows = kv,
-// C cols = q.
+// dK/dV: block = 256 kv rows (8 waves x 32); q-head group loop in-register.
 // ---------------------------------------------------------------------------
-template <int DH>
-__global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
+template <int DH, int WAVES>
+__global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
+    WAVES <= 4 ? 1 : 2))) void fa_bwd_dkv_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dk, float* __restrict__ dv,  // fp32 accum buffers
+    bf16* __restrict__ dk, bf16* __restrict__ dv,
     int S, int H, int HKV, int causal) {
-  constexpr int DC = DH / 32;
+  constexpr int DCH = DH / 16;
+  constexpr int DT = DH / 32;
+  constexpr int KSLOT = DH / 8;
+  constexpr int NT = WAVES * 64;        // threads per block
+  constexpr int KVROWS = WAVES * 32;    // kv rows per block
   const int wave = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int krow0 = blockIdx.x * 64 + wave * 16;  // first KV row
-  const int bh = blockIdx.y;                       // over B*H (q heads!)
-  const int h = bh % H;
-  const int hkv = h / (H / HKV);
-  const int b = bh / H;
-  const bool wave_active = krow0 < S;
-  const bf16* qp = q + (((long)b * H + h) * S) * DH;
+  const int lq = l & 31;
+  const int h2 = l >> 5;
+  const int bh = blockIdx.y;            // over B * HKV
+  const int hkv = bh % HKV;
+  const int b = bh / HKV;
+  const int G = H / HKV;
+  const int kvbase = blockIdx.x * KVROWS;
+  const int kvw0 = kvbase + wave * 32;             // wave's first kv row
+  const int kvrow = min(kvw0 + lq, S - 1);         // lane's kv row (clamped)
+
   const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
   const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
-  const bf16* dop = dout + (((long)b * H + h) * S) * DH;
-  const float* lp = lse + ((long)b * H + h) * S;
-  const float* dp = delta + ((long)b * H + h) * S;
-  // dK/dV accumulate per (b, q-head): summed over q-head groups on the
-  // host for GQA.  Layout: [B, H, S, DH] fp32.
-  float* dkp = dk + (((long)b * H + h) * S) * DH;
-  float* dvp = dv + (((long)b * H + h) * S) * DH;
 
-  // K and V fragments for this wave's 16 kv rows (A-layout: row = l&15)
-  const int krow = min(krow0 + (l & 15), S - 1);
-  bf16x8v kfrag[DC], vfrag[DC];
+  // Wave-private K/V fragments (the 32x32x16 A and B lane maps are
+  // identical, so these registers serve as the B operand directly).
+  // D=128 runs 4 waves x waves_per_eu(1) so each wave owns the whole
+  // 512-VGPR file — resident K/V + 128 accumulator registers, no spill.
+  bf16x8v kf[DCH], vf[DCH];
   #pragma unroll
-  for (int c = 0; c < DC; ++c) {
-    kfrag[c] = *reinterpret_cast<const bf16x8v*>(
-        kp + (long)krow * DH + c * 32 + (l >> 4) * 8);
-    vfrag[c] = *reinterpret_cast<const bf16x8v*>(
-        vp + (long)krow * DH + c * 32 + (l >> 4) * 8);
+  for (int ch = 0; ch < DCH; ++ch) {
+    kf[ch] = *reinterpret_cast<const bf16x8v*>(
+        kp + (long)kvrow * DH + ch * 16 + h2 * 8);
+    vf[ch] = *reinterpret_cast<const bf16x8v*>(
+        vp + (long)kvrow * DH + ch * 16 + h2 * 8);
   }
-  f32x4v dkacc[DH / 16], dvacc[DH / 16];
+
+  f32x16v dkacc[DT], dvacc[DT];
   #pragma unroll
-  for (int f = 0; f < DH / 16; ++f) {
-    dkacc[f] = {0.f, 0.f, 0.f, 0.f};
-    dvacc[f] = {0.f, 0.f, 0.f, 0.f};
-  }
-  __shared__ bf16 st_lds[4][16][32];  // staged P^T / dS^T tiles per wave
+  for (int dt = 0; dt < DT; ++dt)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dkacc[dt][r] = 0.f;
+      dvacc[dt][r] = 0.f;
+    }
+
+  // LDS images of the current q tile
+  __shared__ bf16 qrm[FB_QB * DH];                 // row-major, swizzled
+  __shared__ bf16 dorm[FB_QB * DH];
+  __shared__ bf16 qtr[DH * (FB_QB + FB_PAD)];      // transposed, padded
+  __shared__ bf16 dotr[DH * (FB_QB + FB_PAD)];
+  __shared__ float lsh[FB_QB], dsh[FB_QB];
 
   const float scale = rsqrtf((float)DH);
-  // causal: q tiles start at the block's first kv row
-  const int q0 = causal ? ((blockIdx.x * 64) & ~31) : 0;
-  for (int qb = q0; qb < S; qb += 32) {
-    // ---- S^T quadrants: rows = kv (this wave's 16), cols = q tile half
-    float pt[2][4];   // P^T values for this lane's 4 C rows
-    float dpt[2][4];  // dP^T values
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd) {
-      const int qcol = min(qb + qd * 16 + (l & 15), S - 1);
-      f32x4v sacc = {0.f, 0.f, 0.f, 0.f};
-      f32x4v dpacc = {0.f, 0.f, 0.f, 0.f};
-      #pragma unroll
-      for (int c = 0; c < DC; ++c) {
-        // B-frag from Q rows (S^T = K Q^T) and dO rows (dP^T = V dO^T)
-        bf16x8v qb_frag = *reinterpret_cast<const bf16x8v*>(
-            qp + (long)qcol * DH + c * 32 + (l >> 4) * 8);
-        bf16x8v dob_frag = *reinterpret_cast<const bf16x8v*>(
-            dop + (long)qcol * DH + c * 32 + (l >> 4) * 8);
-        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[c], qb_frag,
-                                                       sacc, 0, 0, 0);
-        dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[c], dob_frag,
-                                                        dpacc, 0, 0, 0);
+  const int tid = threadIdx.x;
+
+  const int q0 = causal ? min(kvbase & ~(FB_QB - 1), S) : 0;
+  for (int gh = 0; gh < G; ++gh) {
+    const int h = hkv * G + gh;
+    const bf16* qp = q + (((long)b * H + h) * S) * DH;
+    const bf16* dop = dout + (((long)b * H + h) * S) * DH;
+    const float* lp = lse + ((long)b * H + h) * S;
+    const float* dp = delta + ((long)b * H + h) * S;
+
+    for (int qb = q0; qb < S; qb += FB_QB) {
+      // ---- stage the q tile (all threads; generic strided loops)
+      __syncthreads();
+      for (int p = tid; p < 2 * FB_QB * KSLOT; p += NT) {
+        const int img = p >= FB_QB * KSLOT;
+        const int pp = p - img * FB_QB * KSLOT;
+        const int row = pp / KSLOT;
+        const int slot = pp % KSLOT;
+        const int grow = min(qb + row, S - 1);
+        const int sslot = slot ^ (row & (KSLOT - 1));
+        const bf16* src = img == 0 ? qp : dop;
+        bf16* dst = img == 0 ? qrm : dorm;
+        *reinterpret_cast<bf16x8v*>(&dst[row * DH + sslot * 8]) =
+            *reinterpret_cast<const bf16x8v*>(src + (long)grow * DH + slot * 8);
       }
-      const float L = lp[min(qb + qd * 16 + (l & 15), S - 1)];
-      const float dl = dp[min(qb + qd * 16 + (l & 15), S - 1)];
+      for (int a = tid; a < 2 * (DH / 8) * 16; a += NT) {
+        const int img = a >= (DH / 8) * 16;
+        const int aa = a - img * (DH / 8) * 16;
+        const int dchunk = aa >> 4;
+        const int kvq = aa & 15;
+        const bf16* src = img == 0 ? qp : dop;
+        bf16* dst = img == 0 ? qtr : dotr;
+        union { bf16x8v v8[4]; short sh[4][8]; } u;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int grow = min(qb + kvq * 4 + i, S - 1);
+          u.v8[i] = *reinterpret_cast<const bf16x8v*>(
+              src + (long)grow * DH + dchunk * 8);
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          union { short s4[4]; unsigned long long d; } pack;
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) pack.s4[i] = u.sh[i][j];
+          *reinterpret_cast<unsigned long long*>(
+              &dst[(dchunk * 8 + j) * (FB_QB + FB_PAD) + kvq * 4]) = pack.d;
+        }
+      }
+      if (tid < FB_QB) {
+        const int grow = min(qb + tid, S - 1);
+        lsh[tid] = lp[grow];
+        dsh[tid] = dp[grow];
+      }
+      __syncthreads();
+
+      // ---- per 32-row q sub-tile: C[q, kv] products, elementwise, repack,
+      // dV^T/dK^T accumulate
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int kr = krow0 + (l >> 4) * 4 + r;  // kv row of this C elem
-        const int qc = qb + qd * 16 + (l & 15);   // q col
-        float p = 0.f;
-        if (qc < S && kr < S && (!causal || kr <= qc))
-          p = __expf(sacc[r] * scale - L);
-        pt[qd][r] = p;
-        dpt[qd][r] = p * (dpacc[r] - dl) * scale;  // = dS^T element
+      for (int ct = 0; ct < 2; ++ct) {
+        f32x16v sacc, dpacc;
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          sacc[r] = 0.f;
+          dpacc[r] = 0.f;
+        }
+        #pragma unroll
+        for (int ch = 0; ch < DCH; ++ch) {
+          const int row = ct * 32 + lq;
+          const int slot = (ch * 2 + h2) ^ (row & (KSLOT - 1));
+          bf16x8v qa = *reinterpret_cast<const bf16x8v*>(
+              &qrm[row * DH + slot * 8]);
+          bf16x8v doa = *reinterpret_cast<const bf16x8v*>(
+              &dorm[row * DH + slot * 8]);
+          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[ch], sacc,
+                                                         0, 0, 0);
+          dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vf[ch], dpacc,
+                                                          0, 0, 0);
+        }
+        // overwrite sacc/dpacc in place with P / dS (register economy)
+        const int kvg = kvw0 + lq;                 // lane's kv column
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qloc = ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+          const int qg = qb + qloc;
+          float p = 0.f;
+          if (qg < S && kvg < S && (!causal || kvg <= qg))
+            p = __expf(sacc[r] * scale - lsh[qloc]);
+          const float ds = p * (dpacc[r] - dsh[qloc]) * scale;
+          sacc[r] = p;
+          dpacc[r] = ds;
+        }
+        bf16x8v pb[2], dsb[2];
+        fb_repack16(sacc, h2, pb);
+        fb_repack16(dpacc, h2, dsb);
+        #pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          #pragma unroll
+          for (int kc = 0; kc < 2; ++kc) {
+            const int qc = (ct * 2 + kc) * 16 + h2 * 8;  // q k-chunk base
+            bf16x8v dota = *reinterpret_cast<const bf16x8v*>(
+                &dotr[(dt * 32 + lq) * (FB_QB + FB_PAD) + qc]);
+            bf16x8v qta = *reinterpret_cast<const bf16x8v*>(
+                &qtr[(dt * 32 + lq) * (FB_QB + FB_PAD) + qc]);
+            dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                dota, pb[kc], dvacc[dt], 0, 0, 0);
+            dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                qta, dsb[kc], dkacc[dt], 0, 0, 0);
+          }
+        }
       }
     }
-    // ---- dV += P^T x dO : stage P^T, MFMA against dO rows
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd)
-      #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        st_lds[wave][(l >> 4) * 4 + r][qd * 16 + (l & 15)] =
-            fb_f2bf(pt[qd][r]);
-    __syncthreads();
-    bf16x8v afrag;
-    #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      afrag[e] = (short)*reinterpret_cast<unsigned short*>(
-          &st_lds[wave][l & 15][(l >> 4) * 8 + e]);
-    #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
-      bf16x8v bfrag;
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int qr = min(qb + (l >> 4) * 8 + e, S - 1);
-        bfrag[e] = (short)*reinterpret_cast<const unsigned short*>(
-            &dop[(long)qr * DH + f * 16 + (l & 15)]);
-      }
-      dvacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                         dvacc[f], 0, 0, 0);
-    }
-    __syncthreads();
-    // ---- dK += dS^T x Q : stage dS^T, MFMA against Q rows
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd)
-      #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        st_lds[wave][(l >> 4) * 4 + r][qd * 16 + (l & 15)] =
-            fb_f2bf(dpt[qd][r]);
-    __syncthreads();
-    #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      afrag[e] = (short)*reinterpret_cast<unsigned short*>(
-          &st_lds[wave][l & 15][(l >> 4) * 8 + e]);
-    #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
-      bf16x8v bfrag;
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int qr = min(qb + (l >> 4) * 8 + e, S - 1);
-        bfrag[e] = (short)*reinterpret_cast<const unsigned short*>(
-            &qp[(long)qr * DH + f * 16 + (l & 15)]);
-      }
-      dkacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                         dkacc[f], 0, 0, 0);
-    }
-    __syncthreads();
   }
-  // epilogue: C rows = kv rows
-  #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int kr = krow0 + (l >> 4) * 4 + r;
-    if (!wave_active || kr >= S) continue;
+
+  // ---- epilogue: C[d, kv] (lane = kv column) -> dk/dv[kv][d] bf16
+  const int kvg = kvw0 + lq;
+  if (kvg < S) {
+    bf16* dkp = dk + (((long)b * HKV + hkv) * S + kvg) * DH;
+    bf16* dvp = dv + (((long)b * HKV + hkv) * S + kvg) * DH;
     #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
-      dkp[(long)kr * DH + f * 16 + (l & 15)] = dkacc[f][r];
-      dvp[(long)kr * DH + f * 16 + (l & 15)] = dvacc[f][r];
+    for (int dt = 0; dt < DT; ++dt) {
+      #pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+        *reinterpret_cast<unsigned int*>(dkp + d) =
+            fb_pk(dkacc[dt][r], dkacc[dt][r + 1]);
+        *reinterpret_cast<unsigned int*>(dvp + d) =
+            fb_pk(dvacc[dt][r], dvacc[dt][r + 1]);
+      }
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-// dQ kernel: wave owns 16 q rows; iterates KV tiles of 32 up to the causal
-// bound.  Mirrors the forward's structure with the dS x K product.
+// dQ: block = 256 q rows (8 waves x 32); mirrors the forward's structure.
 // ---------------------------------------------------------------------------
 template <int DH>
-__global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
+__global__ __launch_bounds__(512) void fa_bwd_dq_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, int S, int H, int HKV, int causal) {
-  constexpr int DC = DH / 32;
+  constexpr int DCH = DH / 16;
+  constexpr int DT = DH / 32;
+  constexpr int KSLOT = DH / 8;
   const int wave = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int row0 = blockIdx.x * 64 + wave * 16;
+  const int lq = l & 31;
+  const int h2 = l >> 5;
   const int bh = blockIdx.y;
   const int h = bh % H;
   const int hkv = h / (H / HKV);
   const int b = bh / H;
-  const bool wave_active = row0 < S;
+  const int qbase = blockIdx.x * 256;
+  const int row0 = qbase + wave * 32;
+  const int qrow = min(row0 + lq, S - 1);
+
   const bf16* qp = q + (((long)b * H + h) * S) * DH;
   const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
   const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
   const bf16* dop = dout + (((long)b * H + h) * S) * DH;
-  const float* lp = lse + ((long)b * H + h) * S;
-  const float* dp = delta + ((long)b * H + h) * S;
+  const float Lq = lse[((long)b * H + h) * S + qrow];
+  const float Dq = delta[((long)b * H + h) * S + qrow];
 
-  const int qrow = min(row0 + (l & 15), S - 1);
-  bf16x8v qfrag[DC], dofrag[DC];
+  bf16x8v qf[DCH], dof[DCH];
   #pragma unroll
-  for (int c = 0; c < DC; ++c) {
-    qfrag[c] = *reinterpret_cast<const bf16x8v*>(
-        qp + (long)qrow * DH + c * 32 + (l >> 4) * 8);
-    dofrag[c] = *reinterpret_cast<const bf16x8v*>(
-        dop + (long)qrow * DH + c * 32 + (l >> 4) * 8);
+  for (int ch = 0; ch < DCH; ++ch) {
+    qf[ch] = *reinterpret_cast<const bf16x8v*>(
+        qp + (long)qrow * DH + ch * 16 + h2 * 8);
+    dof[ch] = *reinterpret_cast<const bf16x8v*>(
+        dop + (long)qrow * DH + ch * 16 + h2 * 8);
   }
-  f32x4v dqacc[DH / 16];
+  f32x16v dqacc[DT];
   #pragma unroll
-  for (int f = 0; f < DH / 16; ++f) dqacc[f] = {0.f, 0.f, 0.f, 0.f};
-  __shared__ bf16 ds_lds[4][16][32];
+  for (int dt = 0; dt < DT; ++dt)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) dqacc[dt][r] = 0.f;
+
+  __shared__ bf16 krm[FB_QB * DH];
+  __shared__ bf16 vrm[FB_QB * DH];
+  __shared__ bf16 ktr[DH * (FB_QB + FB_PAD)];
 
   const float scale = rsqrtf((float)DH);
-  const int block_last_row = min(blockIdx.x * 64 + 63, S - 1);
-  const int kv_end = causal ? (block_last_row + 1) : S;
-  for (int kb = 0; kb < kv_end; kb += 32) {
-    float dsv[2][4];
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd) {
-      const int kcol = min(kb + qd * 16 + (l & 15), S - 1);
-      f32x4v sacc = {0.f, 0.f, 0.f, 0.f};
-      f32x4v dpacc = {0.f, 0.f, 0.f, 0.f};
+  const int tid = threadIdx.x;
+  const bool is_rm = tid < 256;
+  const int ka = tid & 255;
+  constexpr int RPIECE = FB_QB * KSLOT / 256;
+  const int v_dchunk = ka >> 4;
+  const int v_kvq = ka & 15;
+  const bool tr_active = !is_rm && v_dchunk < DH / 8;
+
+  const int kv_end = causal ? min(qbase + 256, S) : S;
+  for (int kb = 0; kb < kv_end; kb += FB_QB) {
+    __syncthreads();
+    if (is_rm) {
       #pragma unroll
-      for (int c = 0; c < DC; ++c) {
-        bf16x8v kb_frag = *reinterpret_cast<const bf16x8v*>(
-            kp + (long)kcol * DH + c * 32 + (l >> 4) * 8);
-        bf16x8v vb_frag = *reinterpret_cast<const bf16x8v*>(
-            vp + (long)kcol * DH + c * 32 + (l >> 4) * 8);
-        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], kb_frag,
-                                                       sacc, 0, 0, 0);
-        dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[c], vb_frag,
-                                                        dpacc, 0, 0, 0);
+      for (int i = 0; i < RPIECE; ++i) {
+        const int p = ka * RPIECE + i;
+        const int row = p / KSLOT;
+        const int slot = p % KSLOT;
+        const int grow = min(kb + row, S - 1);
+        const int sslot = slot ^ (row & (KSLOT - 1));
+        *reinterpret_cast<bf16x8v*>(&krm[row * DH + sslot * 8]) =
+            *reinterpret_cast<const bf16x8v*>(
+                kp + (long)grow * DH + slot * 8);
+        *reinterpret_cast<bf16x8v*>(&vrm[row * DH + sslot * 8]) =
+            *reinterpret_cast<const bf16x8v*>(
+                vp + (long)grow * DH + slot * 8);
+      }
+    } else if (tr_active) {
+      union { bf16x8v v8[4]; short s[4][8]; } u;
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int grow = min(kb + v_kvq * 4 + i, S - 1);
+        u.v8[i] = *reinterpret_cast<const bf16x8v*>(
+            kp + (long)grow * DH + v_dchunk * 8);
       }
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int rr = row0 + (l >> 4) * 4 + r;
-        const int cc = kb + qd * 16 + (l & 15);
+      for (int j = 0; j < 8; ++j) {
+        union { short s4[4]; unsigned long long d; } pack;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) pack.s4[i] = u.s[i][j];
+        *reinterpret_cast<unsigned long long*>(
+            &ktr[(v_dchunk * 8 + j) * (FB_QB + FB_PAD) + v_kvq * 4]) =
+            pack.d;
+      }
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ct = 0; ct < 2; ++ct) {
+      f32x16v sacc, dpacc;
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sacc[r] = 0.f;
+        dpacc[r] = 0.f;
+      }
+      #pragma unroll
+      for (int ch = 0; ch < DCH; ++ch) {
+        const int row = ct * 32 + lq;
+        const int slot = (ch * 2 + h2) ^ (row & (KSLOT - 1));
+        bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(
+            &krm[row * DH + slot * 8]);
+        bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(
+            &vrm[row * DH + slot * 8]);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[ch], sacc,
+                                                       0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ch], dpacc,
+                                                        0, 0, 0);
+      }
+      // C[kv, q]: lane = q column; elementwise uses the lane's own L/delta
+      const int qg = row0 + lq;
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvg = kb + ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
         float p = 0.f;
-        if (cc < S && rr < S && (!causal || cc <= rr))
-          p = __expf(sacc[r] * scale - lp[min(rr, S - 1)]);
-        dsv[qd][r] = p * (dpacc[r] - dp[min(rr, S - 1)]) * scale;
+        if (kvg < S && qg < S && (!causal || kvg <= qg))
+          p = __expf(sacc[r] * scale - Lq);
+        dpacc[r] = p * (dpacc[r] - Dq) * scale;
+      }
+      bf16x8v dsb[2];
+      fb_repack16(dpacc, h2, dsb);
+      #pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        #pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          const int kvc = (ct * 2 + kc) * 16 + h2 * 8;
+          bf16x8v kta = *reinterpret_cast<const bf16x8v*>(
+              &ktr[(dt * 32 + lq) * (FB_QB + FB_PAD) + kvc]);
+          dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kta, dsb[kc], dqacc[dt], 0, 0, 0);
+        }
       }
     }
-    // stage dS, then dQ += dS x K
-    #pragma unroll
-    for (int qd = 0; qd < 2; ++qd)
-      #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        ds_lds[wave][(l >> 4) * 4 + r][qd * 16 + (l & 15)] =
-            fb_f2bf(dsv[qd][r]);
-    __syncthreads();
-    bf16x8v afrag;
-    #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      afrag[e] = (short)*reinterpret_cast<unsigned short*>(
-          &ds_lds[wave][l & 15][(l >> 4) * 8 + e]);
-    #pragma unroll
-    for (int f = 0; f < DH / 16; ++f) {
-      bf16x8v bfrag;
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int kr = min(kb + (l >> 4) * 8 + e, S - 1);
-        bfrag[e] = (short)*reinterpret_cast<const unsigned short*>(
-            &kp[(long)kr * DH + f * 16 + (l & 15)]);
-      }
-      dqacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                         dqacc[f], 0, 0, 0);
-    }
-    __syncthreads();
   }
-  #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int rr = row0 + (l >> 4) * 4 + r;
-    if (!wave_active || rr >= S) continue;
+
+  // epilogue: dQ^T[d, q] (lane = q) -> dq[q][d]
+  const int qg = row0 + lq;
+  if (qg < S) {
+    bf16* dqp = dq + (((long)b * H + h) * S + qg) * DH;
     #pragma unroll
-    for (int f = 0; f < DH / 16; ++f)
-      dq[(((long)b * H + h) * S + rr) * DH + f * 16 + (l & 15)] =
-          fb_f2bf(dqacc[f][r]);
+    for (int dt = 0; dt < DT; ++dt) {
+      #pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+        *reinterpret_cast<unsigned int*>(dqp + d) =
+            fb_pk(dqacc[dt][r], dqacc[dt][r + 1]);
+      }
+    }
   }
 }
 
@@ -318,7 +467,7 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                at::Tensor lse, bool causal) {
   const int B = q.size(0), H = q.size(1), S = q.size(2), DH = q.size(3);
   const int HKV = k.size(1);
-  TORCH_CHECK(DH == 64 || DH == 128, "fa_bwd v0: head dim 64 or 128");
+  TORCH_CHECK(DH == 64 || DH == 128, "fa_bwd: head dim 64 or 128");
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto delta = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   const long rows = (long)B * H * S;
@@ -327,37 +476,36 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                      (const bf16*)out.data_ptr(), delta.data_ptr<float>(),
                      rows, DH);
   auto dq = at::empty_like(q);
-  // per-q-head fp32 dk/dv; GQA groups summed below
-  auto dk_full = at::empty({B, H, S, DH}, q.options().dtype(at::kFloat));
-  auto dv_full = at::empty({B, H, S, DH}, q.options().dtype(at::kFloat));
-  dim3 grid((S + 63) / 64, B * H);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  // D=128 dkv runs 4 waves/block (whole VGPR file per wave: spill-free)
+  dim3 grid_kv128((S + 127) / 128, B * HKV);
+  dim3 grid_kv((S + 255) / 256, B * HKV);
+  dim3 grid_q((S + 255) / 256, B * H);
   if (DH == 128) {
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<128>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_bwd_dkv_v1<128, 4>), grid_kv128, dim3(256), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dk_full.data_ptr<float>(), dv_full.data_ptr<float>(),
+                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
                        S, H, HKV, causal ? 1 : 0);
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<128>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_bwd_dq_v1<128>), grid_q, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), S, H, HKV, causal ? 1 : 0);
   } else {
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<64>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_bwd_dkv_v1<64, 8>), grid_kv, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dk_full.data_ptr<float>(), dv_full.data_ptr<float>(),
+                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
                        S, H, HKV, causal ? 1 : 0);
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<64>), grid, dim3(256), 0, stream,
+    hipLaunchKernelGGL((fa_bwd_dq_v1<64>), grid_q, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (bf16*)dq.data_ptr(), S, H, HKV, causal ? 1 : 0);
   }
-  // GQA: sum q-head groups back to the kv heads (torch op; v0 simplicity)
-  auto dk = dk_full.view({B, HKV, H / HKV, S, DH}).sum(2).to(q.dtype());
-  auto dv = dv_full.view({B, HKV, H / HKV, S, DH}).sum(2).to(q.dtype());
   return {dq, dk, dv};
 }
