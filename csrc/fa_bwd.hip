@@ -112,8 +112,7 @@ __global__ __launch_bounds__(256) void fa_delta_kernel(
 // dK/dV: block = 256 kv rows (8 waves x 32); q-head group loop in-register.
 // ---------------------------------------------------------------------------
 template <int DH, int WAVES>
-__global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
-    WAVES <= 4 ? 1 : 2))) void fa_bwd_dkv_v1(
+__global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2))) void fa_bwd_dkv_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -234,6 +233,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
           sacc[r] = 0.f;
           dpacc[r] = 0.f;
         }
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int ch = 0; ch < DCH; ++ch) {
           const int row = ct * 32 + lq;
@@ -247,6 +247,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
           dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vf[ch], dpacc,
                                                           0, 0, 0);
         }
+        __builtin_amdgcn_s_setprio(0);
         // overwrite sacc/dpacc in place with P / dS (register economy)
         const int kvg = kvw0 + lq;                 // lane's kv column
         #pragma unroll
@@ -263,6 +264,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
         bf16x8v pb[2], dsb[2];
         fb_repack16(sacc, h2, pb);
         fb_repack16(dpacc, h2, dsb);
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
           #pragma unroll
@@ -278,6 +280,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(
                 qta, dsb[kc], dkacc[dt], 0, 0, 0);
           }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
     }
   }

@@ -230,6 +230,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
       f32x16v sacc;
       #pragma unroll
       for (int r = 0; r < 16; ++r) sacc[r] = 0.f;
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int ch = 0; ch < DCH; ++ch) {
         // A-frag: K row kv = ct*32 + lq, k-elems d = ch*16 + h2*8 + e
@@ -240,6 +241,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
         sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ch], sacc,
                                                        0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kvg = kb + ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
@@ -314,6 +316,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     }
 
     // ---- O^T += V^T x P^T   (A: V^T[d][kv] contiguous b128 from vbuf)
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       #pragma unroll
@@ -324,6 +327,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
                                                            oacc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // every wave done reading this tile's K/V images
     if (kb + FA_KVB < kv_end) {
       write_tile();
