@@ -94,14 +94,22 @@ __device__ __forceinline__ void fb_repack16(const f32x16v& pr, int h2,
 // delta[b,h,s] = sum_d dO[b,h,s,d] * O[b,h,s,d]  (fp32)
 __global__ __launch_bounds__(256) void fa_delta_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ out,
-    float* __restrict__ delta, long rows, int DH) {
+    float* __restrict__ delta, long rows, int DH, int H, int S,
+    long dsb, long dsh, long dss, long osb, long osh, long oss) {
   const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
   const int l = threadIdx.x & 63;
   if (row >= rows) return;
+  const int sres = (int)(row % S);
+  const int hres = (int)((row / S) % H);
+  const int bres = (int)(row / ((long)S * H));
+  const bf16* dop = dout + (long)bres * dsb + (long)hres * dsh +
+                    (long)sres * dss;
+  const bf16* op = out + (long)bres * osb + (long)hres * osh +
+                   (long)sres * oss;
   float s = 0.f;
   for (int d = l * 2; d < DH; d += 128) {
-    s += fb_b2f(dout[row * DH + d]) * fb_b2f(out[row * DH + d]);
-    s += fb_b2f(dout[row * DH + d + 1]) * fb_b2f(out[row * DH + d + 1]);
+    s += fb_b2f(dop[d]) * fb_b2f(op[d]);
+    s += fb_b2f(dop[d + 1]) * fb_b2f(op[d + 1]);
   }
   #pragma unroll
   for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off);
@@ -117,7 +125,9 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dk, bf16* __restrict__ dv,
-    int S, int H, int HKV, int causal) {
+    int S, int H, int HKV, int causal,
+    long qsb, long qsh, long qss, long ksb, long ksh, long kss,
+    long vsb, long vsh, long vss, long dosb, long dosh, long doss) {
   constexpr int DCH = DH / 16;
   constexpr int DT = DH / 32;
   constexpr int KSLOT = DH / 8;
@@ -135,8 +145,8 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   const int kvw0 = kvbase + wave * 32;             // wave's first kv row
   const int kvrow = min(kvw0 + lq, S - 1);         // lane's kv row (clamped)
 
-  const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
-  const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
+  const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
+  const bf16* vp = v + (long)b * vsb + (long)hkv * vsh;
 
   // Wave-private K/V fragments (the 32x32x16 A and B lane maps are
   // identical, so these registers serve as the B operand directly).
@@ -146,9 +156,9 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   #pragma unroll
   for (int ch = 0; ch < DCH; ++ch) {
     kf[ch] = *reinterpret_cast<const bf16x8v*>(
-        kp + (long)kvrow * DH + ch * 16 + h2 * 8);
+        kp + (long)kvrow * kss + ch * 16 + h2 * 8);
     vf[ch] = *reinterpret_cast<const bf16x8v*>(
-        vp + (long)kvrow * DH + ch * 16 + h2 * 8);
+        vp + (long)kvrow * vss + ch * 16 + h2 * 8);
   }
 
   f32x16v dkacc[DT], dvacc[DT];
@@ -173,8 +183,8 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   const int q0 = causal ? min(kvbase & ~(FB_QB - 1), S) : 0;
   for (int gh = 0; gh < G; ++gh) {
     const int h = hkv * G + gh;
-    const bf16* qp = q + (((long)b * H + h) * S) * DH;
-    const bf16* dop = dout + (((long)b * H + h) * S) * DH;
+    const bf16* qp = q + (long)b * qsb + (long)h * qsh;
+    const bf16* dop = dout + (long)b * dosb + (long)h * dosh;
     const float* lp = lse + ((long)b * H + h) * S;
     const float* dp = delta + ((long)b * H + h) * S;
 
@@ -189,9 +199,10 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
         const int grow = min(qb + row, S - 1);
         const int sslot = slot ^ (row & (KSLOT - 1));
         const bf16* src = img == 0 ? qp : dop;
+        const long rs = img == 0 ? qss : doss;
         bf16* dst = img == 0 ? qrm : dorm;
         *reinterpret_cast<bf16x8v*>(&dst[row * DH + sslot * 8]) =
-            *reinterpret_cast<const bf16x8v*>(src + (long)grow * DH + slot * 8);
+            *reinterpret_cast<const bf16x8v*>(src + (long)grow * rs + slot * 8);
       }
       for (int a = tid; a < 2 * (DH / 8) * 16; a += NT) {
         const int img = a >= (DH / 8) * 16;
@@ -199,13 +210,14 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
         const int dchunk = aa >> 4;
         const int kvq = aa & 15;
         const bf16* src = img == 0 ? qp : dop;
+        const long rs = img == 0 ? qss : doss;
         bf16* dst = img == 0 ? qtr : dotr;
         union { bf16x8v v8[4]; short sh[4][8]; } u;
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int grow = min(qb + kvq * 4 + i, S - 1);
           u.v8[i] = *reinterpret_cast<const bf16x8v*>(
-              src + (long)grow * DH + dchunk * 8);
+              src + (long)grow * rs + dchunk * 8);
         }
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -290,6 +302,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   if (kvg < S) {
     bf16* dkp = dk + (((long)b * HKV + hkv) * S + kvg) * DH;
     bf16* dvp = dv + (((long)b * HKV + hkv) * S + kvg) * DH;
+    // (dk/dv are allocated contiguous [B,HKV,S,D] by the host wrapper)
     #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       #pragma unroll
@@ -312,7 +325,9 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    bf16* __restrict__ dq, int S, int H, int HKV, int causal) {
+    bf16* __restrict__ dq, int S, int H, int HKV, int causal,
+    long qsb, long qsh, long qss, long ksb, long ksh, long kss,
+    long vsb, long vsh, long vss, long dosb, long dosh, long doss) {
   constexpr int DCH = DH / 16;
   constexpr int DT = DH / 32;
   constexpr int KSLOT = DH / 8;
@@ -328,10 +343,10 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
   const int row0 = qbase + wave * 32;
   const int qrow = min(row0 + lq, S - 1);
 
-  const bf16* qp = q + (((long)b * H + h) * S) * DH;
-  const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
-  const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
-  const bf16* dop = dout + (((long)b * H + h) * S) * DH;
+  const bf16* qp = q + (long)b * qsb + (long)h * qsh;
+  const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
+  const bf16* vp = v + (long)b * vsb + (long)hkv * vsh;
+  const bf16* dop = dout + (long)b * dosb + (long)h * dosh;
   const float Lq = lse[((long)b * H + h) * S + qrow];
   const float Dq = delta[((long)b * H + h) * S + qrow];
 
@@ -339,9 +354,9 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
   #pragma unroll
   for (int ch = 0; ch < DCH; ++ch) {
     qf[ch] = *reinterpret_cast<const bf16x8v*>(
-        qp + (long)qrow * DH + ch * 16 + h2 * 8);
+        qp + (long)qrow * qss + ch * 16 + h2 * 8);
     dof[ch] = *reinterpret_cast<const bf16x8v*>(
-        dop + (long)qrow * DH + ch * 16 + h2 * 8);
+        dop + (long)qrow * doss + ch * 16 + h2 * 8);
   }
   f32x16v dqacc[DT];
   #pragma unroll
@@ -375,10 +390,10 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
         const int sslot = slot ^ (row & (KSLOT - 1));
         *reinterpret_cast<bf16x8v*>(&krm[row * DH + sslot * 8]) =
             *reinterpret_cast<const bf16x8v*>(
-                kp + (long)grow * DH + slot * 8);
+                kp + (long)grow * kss + slot * 8);
         *reinterpret_cast<bf16x8v*>(&vrm[row * DH + sslot * 8]) =
             *reinterpret_cast<const bf16x8v*>(
-                vp + (long)grow * DH + slot * 8);
+                vp + (long)grow * vss + slot * 8);
       }
     } else if (tr_active) {
       union { bf16x8v v8[4]; short s[4][8]; } u;
@@ -386,7 +401,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
       for (int i = 0; i < 4; ++i) {
         const int grow = min(kb + v_kvq * 4 + i, S - 1);
         u.v8[i] = *reinterpret_cast<const bf16x8v*>(
-            kp + (long)grow * DH + v_dchunk * 8);
+            kp + (long)grow * kss + v_dchunk * 8);
       }
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -465,9 +480,19 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
 
 }  // namespace
 
+static bool fb_strides_ok(const at::Tensor& t) {
+  return t.stride(3) == 1 && t.stride(0) % 8 == 0 && t.stride(1) % 8 == 0 &&
+         t.stride(2) % 8 == 0;
+}
+
 std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                at::Tensor out, at::Tensor dout,
                                at::Tensor lse, bool causal) {
+  if (!fb_strides_ok(q)) q = q.contiguous();
+  if (!fb_strides_ok(k)) k = k.contiguous();
+  if (!fb_strides_ok(v)) v = v.contiguous();
+  if (!fb_strides_ok(out)) out = out.contiguous();
+  if (!fb_strides_ok(dout)) dout = dout.contiguous();
   const int B = q.size(0), H = q.size(1), S = q.size(2), DH = q.size(3);
   const int HKV = k.size(1);
   TORCH_CHECK(DH == 64 || DH == 128, "fa_bwd: head dim 64 or 128");
@@ -477,38 +502,47 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   hipLaunchKernelGGL(fa_delta_kernel, dim3((unsigned)((rows + 3) / 4)),
                      dim3(256), 0, stream, (const bf16*)dout.data_ptr(),
                      (const bf16*)out.data_ptr(), delta.data_ptr<float>(),
-                     rows, DH);
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+                     rows, DH, H, S,
+                     dout.stride(0), dout.stride(1), dout.stride(2),
+                     out.stride(0), out.stride(1), out.stride(2));
+  // grads allocated contiguous [B,H,S,D]/[B,HKV,S,D] (autograd accepts
+  // any layout; the kernels' write paths assume dense BHSD)
+  auto dq = at::empty({B, H, S, DH}, q.options());
+  auto dk = at::empty({B, HKV, S, DH}, q.options());
+  auto dv = at::empty({B, HKV, S, DH}, q.options());
   // D=128 dkv runs 4 waves/block (whole VGPR file per wave: spill-free)
   dim3 grid_kv128((S + 127) / 128, B * HKV);
   dim3 grid_kv((S + 255) / 256, B * HKV);
   dim3 grid_q((S + 255) / 256, B * H);
+  const auto LKV = [&](auto kern, dim3 g, dim3 blk) {
+    hipLaunchKernelGGL(kern, g, blk, 0, stream,
+                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
+                       S, H, HKV, causal ? 1 : 0,
+                       q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2),
+                       v.stride(0), v.stride(1), v.stride(2),
+                       dout.stride(0), dout.stride(1), dout.stride(2));
+  };
+  const auto LQ = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid_q, dim3(512), 0, stream,
+                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (bf16*)dq.data_ptr(), S, H, HKV, causal ? 1 : 0,
+                       q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2),
+                       v.stride(0), v.stride(1), v.stride(2),
+                       dout.stride(0), dout.stride(1), dout.stride(2));
+  };
   if (DH == 128) {
-    hipLaunchKernelGGL((fa_bwd_dkv_v1<128, 4>), grid_kv128, dim3(256), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
-                       S, H, HKV, causal ? 1 : 0);
-    hipLaunchKernelGGL((fa_bwd_dq_v1<128>), grid_q, dim3(512), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (bf16*)dq.data_ptr(), S, H, HKV, causal ? 1 : 0);
+    LKV(fa_bwd_dkv_v1<128, 4>, grid_kv128, dim3(256));
+    LQ(fa_bwd_dq_v1<128>);
   } else {
-    hipLaunchKernelGGL((fa_bwd_dkv_v1<64, 8>), grid_kv, dim3(512), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
-                       S, H, HKV, causal ? 1 : 0);
-    hipLaunchKernelGGL((fa_bwd_dq_v1<64>), grid_q, dim3(512), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (bf16*)dq.data_ptr(), S, H, HKV, causal ? 1 : 0);
+    LKV(fa_bwd_dkv_v1<64, 8>, grid_kv, dim3(512));
+    LQ(fa_bwd_dq_v1<64>);
   }
   return {dq, dk, dv};
 }

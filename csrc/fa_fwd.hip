@@ -107,11 +107,17 @@ __global__ void mfma_probe_32x32x16(const bf16* __restrict__ A,  // [32,16] rm
 #define FA_KVB 64  // kv rows per staged tile
 #define FA_VPAD 8  // V^T row padding (elements) to spread write banks
 
+// Strided addressing: every tensor is a 4-D view with unit stride on the
+// head dim; (sb, sh, ss) are the batch/head/row strides in ELEMENTS, so
+// both [B,H,S,D]-contiguous and [B,S,H,D]-contiguous (attention's natural
+// projection layout — no transpose copies) run the same kernel.
 template <int DH>
 __global__ __launch_bounds__(512) void fa_fwd_v1(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
-    float* __restrict__ lse, int S, int H, int HKV, int causal) {
+    float* __restrict__ lse, int S, int H, int HKV, int causal,
+    long qsb, long qsh, long qss, long ksb, long ksh, long kss,
+    long vsb, long vsh, long vss, long osb, long osh, long oss) {
   constexpr int DCH = DH / 16;    // 16-wide k-chunks of the head dim
   constexpr int DT = DH / 32;     // 32-row d-tiles of O^T
   constexpr int KSLOT = DH / 8;   // 16-B slots per K row
@@ -127,9 +133,9 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   const int row0 = qbase + wave * 32;          // wave's first q row
   const int qrow = min(row0 + lq, S - 1);      // this lane's q row (clamped)
 
-  const bf16* qp = q + (((long)b * H + h) * S) * DH;
-  const bf16* kp = k + (((long)b * HKV + hkv) * S) * DH;
-  const bf16* vp = v + (((long)b * HKV + hkv) * S) * DH;
+  const bf16* qp = q + (long)b * qsb + (long)h * qsh;
+  const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
+  const bf16* vp = v + (long)b * vsb + (long)hkv * vsh;
 
   // K image: row-major [KVB][DH], 16-B slot index XOR-swizzled with row&15.
   // V image: transposed [DH][KVB + pad].
@@ -142,7 +148,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   #pragma unroll
   for (int ch = 0; ch < DCH; ++ch)
     qf[ch] = *reinterpret_cast<const bf16x8v*>(
-        qp + (long)qrow * DH + ch * 16 + h2 * 8);
+        qp + (long)qrow * qss + ch * 16 + h2 * 8);
 
   f32x16v oacc[DT];
   #pragma unroll
@@ -176,14 +182,14 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
         const int slot = p % KSLOT;
         const int grow = min(kb + row, S - 1);
         stage[i] = *reinterpret_cast<const bf16x8v*>(
-            kp + (long)grow * DH + slot * 8);
+            kp + (long)grow * kss + slot * 8);
       }
     } else if (v_active) {
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int grow = min(kb + v_kvq * 4 + i, S - 1);
         stage[i] = *reinterpret_cast<const bf16x8v*>(
-            vp + (long)grow * DH + v_dchunk * 8);
+            vp + (long)grow * vss + v_dchunk * 8);
       }
     }
   };
@@ -340,7 +346,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   const int qg = row0 + lq;
   if (qg < S) {
     const float inv = 1.f / fmaxf(lsum, 1e-30f);
-    bf16* op = out + (((long)b * H + h) * S + qg) * DH;
+    bf16* op = out + (long)b * osb + (long)h * osh + (long)qg * oss;
     #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       #pragma unroll
@@ -371,29 +377,44 @@ void mfma_probe32(at::Tensor A, at::Tensor B, at::Tensor D) {
                      D.data_ptr<float>());
 }
 
+static bool fa_strides_ok(const at::Tensor& t) {
+  // unit stride on the head dim, 8-element (16-B) alignment on the others
+  return t.stride(3) == 1 && t.stride(0) % 8 == 0 && t.stride(1) % 8 == 0 &&
+         t.stride(2) % 8 == 0;
+}
+
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                bool causal) {
-  TORCH_CHECK(q.scalar_type() == at::kBFloat16 && q.dim() == 4 &&
-                  q.is_contiguous() && k.is_contiguous() && v.is_contiguous(),
-              "fa_fwd: contiguous [B,H,S,D] bf16");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16 && q.dim() == 4,
+              "fa_fwd: 4-D bf16 [B,H,S,D] (any stride, unit head-dim)");
+  if (!fa_strides_ok(q)) q = q.contiguous();
+  if (!fa_strides_ok(k)) k = k.contiguous();
+  if (!fa_strides_ok(v)) v = v.contiguous();
   const int B = q.size(0), H = q.size(1), S = q.size(2), DH = q.size(3);
   const int HKV = k.size(1);
   TORCH_CHECK(DH == 64 || DH == 128, "fa_fwd: head dim 64 or 128");
   TORCH_CHECK(H % HKV == 0, "fa_fwd: H must be a multiple of H_kv");
-  auto out = at::empty_like(q);
+  // out in [B,S,H,D]-contiguous storage, returned as a [B,H,S,D] view:
+  // the caller's transpose(1,2).reshape(B,S,H*D) is then a free view.
+  auto out_bshd = at::empty({B, S, H, DH}, q.options());
+  auto out = out_bshd.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
   dim3 grid((S + 255) / 256, B * H);
+  const auto L = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(512), 0, stream,
+                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+                       (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
+                       lse.data_ptr<float>(), S, H, HKV, causal ? 1 : 0,
+                       q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2),
+                       v.stride(0), v.stride(1), v.stride(2),
+                       out.stride(0), out.stride(1), out.stride(2));
+  };
   if (DH == 128) {
-    hipLaunchKernelGGL((fa_fwd_v1<128>), grid, dim3(512), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
-                       lse.data_ptr<float>(), S, H, HKV, causal ? 1 : 0);
+    L(fa_fwd_v1<128>);
   } else {
-    hipLaunchKernelGGL((fa_fwd_v1<64>), grid, dim3(512), 0, stream,
-                       (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-                       (const bf16*)v.data_ptr(), (bf16*)out.data_ptr(),
-                       lse.data_ptr<float>(), S, H, HKV, causal ? 1 : 0);
+    L(fa_fwd_v1<64>);
   }
   return {out, lse};
 }

@@ -31,16 +31,20 @@ class _FlashAttnFn(torch.autograd.Function):
         from stoke import _C
 
         q, k, v, out, lse = ctx.saved_tensors
-        dq, dk, dv = _C.fa_bwd(q, k, v, out, dout.contiguous(), lse,
-                               ctx.causal)
+        dq, dk, dv = _C.fa_bwd(q, k, v, out, dout, lse, ctx.causal)
         return dq, dk, dv, None
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = True) -> torch.Tensor:
-    """q: [B,H,S,D], k/v: [B,Hkv,S,D] bf16 contiguous; D in {64, 128}."""
-    return _FlashAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                              causal)
+    """q: [B,H,S,D], k/v: [B,Hkv,S,D] bf16, D in {64, 128}.
+
+    Any strides with a unit head dim are handled natively (the kernels
+    take element strides), so transposed projection views — [B,S,H,D]
+    storage — run with ZERO copies; the output is a [B,H,S,D] view of
+    [B,S,H,D] storage, making the caller's transpose+reshape free too.
+    """
+    return _FlashAttnFn.apply(q, k, v, causal)
 
 
 def _fa_usable(q, k, v, causal) -> bool:
