@@ -180,7 +180,81 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   const float scale = rsqrtf((float)DH);
   const int tid = threadIdx.x;
 
+  // ---- tile pipeline (T14): the row-major Q/dO pieces for tile t+1 are
+  // loaded to registers during tile t's compute; the transposed images are
+  // built from the row-major LDS images (halving global traffic).  Three
+  // barriers per tile: [write rm] | [build tr from rm] | [compute].
+  constexpr int NPREF = 2 * FB_QB * KSLOT / NT;  // rm pieces per thread
+  bf16x8v pref[NPREF];
+
+  auto load_rm = [&](const bf16* qp, const bf16* dop, long qss_, long doss_,
+                     int qb) {
+    #pragma unroll
+    for (int i = 0; i < NPREF; ++i) {
+      const int p = tid + i * NT;
+      const int img = p >= FB_QB * KSLOT;
+      const int pp = p - img * FB_QB * KSLOT;
+      const int row = pp / KSLOT;
+      const int slot = pp % KSLOT;
+      const int grow = min(qb + row, S - 1);
+      const bf16* src = img == 0 ? qp : dop;
+      const long rs = img == 0 ? qss_ : doss_;
+      pref[i] = *reinterpret_cast<const bf16x8v*>(
+          src + (long)grow * rs + slot * 8);
+    }
+  };
+
+  auto write_rm = [&]() {
+    #pragma unroll
+    for (int i = 0; i < NPREF; ++i) {
+      const int p = tid + i * NT;
+      const int img = p >= FB_QB * KSLOT;
+      const int pp = p - img * FB_QB * KSLOT;
+      const int row = pp / KSLOT;
+      const int slot = pp % KSLOT;
+      const int sslot = slot ^ (row & (KSLOT - 1));
+      bf16* dst = img == 0 ? qrm : dorm;
+      *reinterpret_cast<bf16x8v*>(&dst[row * DH + sslot * 8]) = pref[i];
+    }
+  };
+
+  auto build_tr = [&]() {
+    for (int a = tid; a < 2 * (DH / 8) * 16; a += NT) {
+      const int img = a >= (DH / 8) * 16;
+      const int aa = a - img * (DH / 8) * 16;
+      const int dchunk = aa >> 4;
+      const int rq = aa & 15;             // 4-row group of q rows
+      const bf16* srcm = img == 0 ? qrm : dorm;
+      bf16* dst = img == 0 ? qtr : dotr;
+      union { bf16x8v v8[4]; short sh[4][8]; } u;
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = rq * 4 + i;
+        const int sslot = dchunk ^ (row & (KSLOT - 1));
+        u.v8[i] = *reinterpret_cast<const bf16x8v*>(
+            &srcm[row * DH + sslot * 8]);
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { short s4[4]; unsigned long long d; } pack;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) pack.s4[i] = u.sh[i][j];
+        *reinterpret_cast<unsigned long long*>(
+            &dst[(dchunk * 8 + j) * (FB_QB + FB_PAD) + rq * 4]) = pack.d;
+      }
+    }
+  };
+
   const int q0 = causal ? min(kvbase & ~(FB_QB - 1), S) : 0;
+  {
+    const int h0 = hkv * G;
+    load_rm(q + (long)b * qsb + (long)h0 * qsh,
+            dout + (long)b * dosb + (long)h0 * dosh, qss, doss, q0);
+  }
+  write_rm();
+  __syncthreads();
+  build_tr();
+
   for (int gh = 0; gh < G; ++gh) {
     const int h = hkv * G + gh;
     const bf16* qp = q + (long)b * qsb + (long)h * qsh;
@@ -189,51 +263,17 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
     const float* dp = delta + ((long)b * H + h) * S;
 
     for (int qb = q0; qb < S; qb += FB_QB) {
-      // ---- stage the q tile (all threads; generic strided loops)
-      __syncthreads();
-      for (int p = tid; p < 2 * FB_QB * KSLOT; p += NT) {
-        const int img = p >= FB_QB * KSLOT;
-        const int pp = p - img * FB_QB * KSLOT;
-        const int row = pp / KSLOT;
-        const int slot = pp % KSLOT;
-        const int grow = min(qb + row, S - 1);
-        const int sslot = slot ^ (row & (KSLOT - 1));
-        const bf16* src = img == 0 ? qp : dop;
-        const long rs = img == 0 ? qss : doss;
-        bf16* dst = img == 0 ? qrm : dorm;
-        *reinterpret_cast<bf16x8v*>(&dst[row * DH + sslot * 8]) =
-            *reinterpret_cast<const bf16x8v*>(src + (long)grow * rs + slot * 8);
-      }
-      for (int a = tid; a < 2 * (DH / 8) * 16; a += NT) {
-        const int img = a >= (DH / 8) * 16;
-        const int aa = a - img * (DH / 8) * 16;
-        const int dchunk = aa >> 4;
-        const int kvq = aa & 15;
-        const bf16* src = img == 0 ? qp : dop;
-        const long rs = img == 0 ? qss : doss;
-        bf16* dst = img == 0 ? qtr : dotr;
-        union { bf16x8v v8[4]; short sh[4][8]; } u;
-        #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int grow = min(qb + kvq * 4 + i, S - 1);
-          u.v8[i] = *reinterpret_cast<const bf16x8v*>(
-              src + (long)grow * rs + dchunk * 8);
-        }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          union { short s4[4]; unsigned long long d; } pack;
-          #pragma unroll
-          for (int i = 0; i < 4; ++i) pack.s4[i] = u.sh[i][j];
-          *reinterpret_cast<unsigned long long*>(
-              &dst[(dchunk * 8 + j) * (FB_QB + FB_PAD) + kvq * 4]) = pack.d;
-        }
-      }
       if (tid < FB_QB) {
         const int grow = min(qb + tid, S - 1);
         lsh[tid] = lp[grow];
         dsh[tid] = dp[grow];
       }
       __syncthreads();
+
+      // next tile in the (gh, qb) sequence, for the prefetch
+      const int qb_n = qb + FB_QB < S ? qb + FB_QB : q0;
+      const int gh_n = qb + FB_QB < S ? gh : gh + 1;
+      const bool has_next = gh_n < G;
 
       // ---- per 32-row q sub-tile: C[q, kv] products, elementwise, repack,
       // dV^T/dK^T accumulate
@@ -260,6 +300,12 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
                                                           0, 0, 0);
         }
         __builtin_amdgcn_s_setprio(0);
+        if (ct == 0 && has_next) {
+          // issue next tile's global loads under the remaining compute
+          const int hh = hkv * G + gh_n;
+          load_rm(q + (long)b * qsb + (long)hh * qsh,
+                  dout + (long)b * dosb + (long)hh * dosh, qss, doss, qb_n);
+        }
         // overwrite sacc/dpacc in place with P / dS (register economy)
         const int kvg = kvw0 + lq;                 // lane's kv column
         #pragma unroll
@@ -293,6 +339,12 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
           }
         }
         __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();              // everyone done reading rm + tr
+      if (has_next) {
+        write_rm();
+        __syncthreads();            // rm image of tile t+1 complete
+        build_tr();
       }
     }
   }
