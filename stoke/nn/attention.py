@@ -48,10 +48,14 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 def _fa_usable(q, k, v, causal) -> bool:
-    # Opt-in while the hand-written kernels trail AOTriton: v0 measured
-    # 141.7k vs 202.9k tok/s on the GPT-2 bench (gpurun_out/call2.log).
-    # Flip the default once fa_bench shows parity or better.
-    if os.environ.get("STOKE_USE_FA", "0") != "1":
+    # Default policy from measurement (benchmarks/fa_bench.py on MI355X):
+    # D=64 f+b beats AOTriton (1.07x) -> on by default; D=128 trails
+    # (0.84x) -> SDPA unless forced.  STOKE_USE_FA=1 forces the native
+    # kernels everywhere, =0 disables them.
+    mode = os.environ.get("STOKE_USE_FA", "auto")
+    if mode == "0":
+        return False
+    if mode != "1" and q.shape[-1] != 64:
         return False
     if not (q.is_cuda and q.dtype == torch.bfloat16
             and k.dtype == torch.bfloat16 and v.dtype == torch.bfloat16):
