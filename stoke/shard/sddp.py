@@ -26,9 +26,11 @@ from stoke.shard.oss import OSSOptimizer
 
 class _OwnerBucket:
     __slots__ = ("owner", "params", "numel", "ready", "launched", "flat",
-                 "comm_flat", "work")
+                 "comm_flat", "comm_persist", "prediv", "work")
 
     def __init__(self, owner: int):
+        self.comm_persist = None
+        self.prediv = False
         self.owner = owner
         self.params: List[torch.nn.Parameter] = []
         self.numel = 0
@@ -109,9 +111,11 @@ class StokeSDDPModule(torch.nn.Module):
 
     def _launch(self, b: _OwnerBucket):
         b.launched = True
-        flat = torch.empty(
-            b.numel, dtype=b.params[0].dtype, device=b.params[0].device
-        )
+        if b.flat is None:  # persistent pack buffer (reused every step)
+            b.flat = torch.empty(
+                b.numel, dtype=b.params[0].dtype, device=b.params[0].device
+            )
+        flat = b.flat
         offset = 0
         for p in b.params:
             n = p.numel()
@@ -121,10 +125,19 @@ class StokeSDDPModule(torch.nn.Module):
             else:
                 dst.copy_(p.grad.reshape(-1))
             offset += n
-        b.flat = flat
         comm = flat
         if self._reduce_fp16 and flat.dtype == torch.float32:
-            comm = flat.to(torch.float16)
+            if b.comm_persist is None:
+                b.comm_persist = torch.empty(
+                    b.numel, dtype=torch.float16, device=flat.device
+                )
+            comm = b.comm_persist
+            # Pre-divide by world size BEFORE the fp16 SUM so partial sums
+            # stay in fp16 range at world 8
+            torch.mul(flat, 1.0 / self._pg.world_size, out=comm)
+            b.prediv = True
+        else:
+            b.prediv = False
         b.comm_flat = comm
         b.work = dist.reduce(comm, dst=b.owner, async_op=True)
 
@@ -144,7 +157,8 @@ class StokeSDDPModule(torch.nn.Module):
                 b.work = None
             if b.owner == self._pg.rank:
                 comm = b.comm_flat
-                comm.mul_(inv_w)
+                if not b.prediv:
+                    comm.mul_(inv_w)
                 if comm is not b.flat:
                     b.flat.copy_(comm)
                 offset = 0
@@ -159,7 +173,6 @@ class StokeSDDPModule(torch.nn.Module):
                 # Not the owner: gradient shard lives elsewhere; free ours.
                 for p in b.params:
                     p.grad = None
-            b.flat = None
             b.comm_flat = None
             b.ready = 0
             b.launched = False

@@ -256,3 +256,44 @@ def test_fsdp_checkpoint_world2_loads_world1(tmp_path):
         _fsdp_load_w1_worker, args=(1, free_port(), str(tmp_path)), nprocs=1,
         join=True,
     )
+
+
+# ------------------------------------------------- SDDP fp16 reduce safety
+def _sddp_fp16_worker(rank, world, port):
+    import torch.nn as nn
+    from tests.conftest import init_gloo
+    from stoke.shard import OSSOptimizer, StokeSDDPModule
+
+    pg = init_gloo(rank, world, port)
+    torch.manual_seed(0)
+    model = nn.Linear(8, 8, bias=False)
+    opt = OSSOptimizer([p for p in model.parameters()],
+                       optim=torch.optim.SGD, pg=pg, lr=0.0)
+    sddp = StokeSDDPModule(model, sharded_optimizer=opt, pg=pg,
+                           reduce_fp16=True)
+    # Large-magnitude grads: the raw fp16 SUM over ranks would overflow
+    # (2 x 40000 > 65504); the pre-divide keeps partials in range.
+    x = torch.full((4, 8), 100.0)
+    out = sddp(x)
+    (out.sum() * 100.0).backward()
+    sddp.finish_backward()
+    owned = [p.grad for p in model.parameters() if p.grad is not None]
+    if rank == sddp._buckets[0].owner:
+        assert owned and all(torch.isfinite(g).all() for g in owned), \
+            "fp16 reduce overflowed despite pre-divide"
+        # mean semantics preserved: both ranks saw identical data
+        ref = torch.autograd.grad(
+            (model(x).sum() * 100.0), model.parameters()
+        )
+        for g, r in zip(owned, ref):
+            assert torch.allclose(g.float(), r.float(), rtol=2e-2), \
+                (g - r).abs().max()
+    torch.distributed.destroy_process_group()
+
+
+def test_sddp_fp16_reduce_prediv_no_overflow():
+    from tests.conftest import free_port
+
+    torch.multiprocessing.spawn(
+        _sddp_fp16_worker, args=(2, free_port()), nprocs=2, join=True
+    )
