@@ -61,16 +61,22 @@ class SyntheticDocs(torch.utils.data.Dataset):
 
 
 def pad_collate(batch):
+    # inputs pad with 0; TARGETS pad with -100 so padded positions are
+    # ignored by the loss instead of trained as real next tokens
+    # 
     L = max(t.numel() for t in batch)
     x = torch.zeros(len(batch), L, dtype=torch.long)
+    y = torch.full((len(batch), L), -100, dtype=torch.long)
     for i, t in enumerate(batch):
         x[i, : t.numel()] = t
-    return x[:, :-1], x[:, 1:]  # next-token prediction
+        y[i, : t.numel()] = t
+    return x[:, :-1], y[:, 1:]  # next-token prediction
 
 
 def lm_loss(logits, target):
     return nn.functional.cross_entropy(
-        logits.reshape(-1, logits.shape[-1]).float(), target.reshape(-1)
+        logits.reshape(-1, logits.shape[-1]).float(), target.reshape(-1),
+        ignore_index=-100,
     )
 
 

@@ -65,4 +65,7 @@ __all__ = [
     "BucketedDistributedSampler",
 ]
 
-__version__ = "0.1.0"
+from stoke._version import get_versions as _get_versions
+
+__version__ = _get_versions()["version"]
+del _get_versions

@@ -1,1 +1,1 @@
-from stoke.amp.scaler import StokeGradScaler  # noqa: F401
+from stoke.amp.scaler import StokeGradScaler, StokePerLossScaler  # noqa: F401

@@ -196,3 +196,101 @@ class StokeGradScaler:
         self._backoff_factor = state_dict["backoff_factor"]
         self._growth_interval = state_dict["growth_interval"]
         self._growth_tracker.fill_(state_dict["_growth_tracker"])
+
+
+class StokePerLossScaler(StokeGradScaler):
+    """Per-loss dynamic scalers (apex ``scaler_per_loss`` semantics,
+    reference ``fp16.py:545-579`` + ``ApexConfig.scaler_per_loss``).
+
+    Each loss index owns an independent scale/growth tracker.  Every scaled
+    backward is immediately unscaled — the per-loss gradient contribution is
+    isolated with a stash — so a NaN-prone loss backs off only its own scale
+    and step-time gradients are already in true units (``unscale_`` becomes
+    a no-op; the step decision uses the OR of the per-loss inf flags).
+    """
+
+    def __init__(self, **kw):
+        super().__init__(**kw)
+        self._loss_scales: List[torch.Tensor] = []
+        self._loss_trackers: List[torch.Tensor] = []
+        self._loss_found: List[torch.Tensor] = []
+
+    def _loss_state(self, idx: int) -> torch.Tensor:
+        self._lazy_init()
+        while len(self._loss_scales) <= idx:
+            self._loss_scales.append(
+                torch.full((1,), float(self._init_scale), device=self._device)
+            )
+            self._loss_trackers.append(
+                torch.zeros(1, dtype=torch.int32, device=self._device)
+            )
+            self._loss_found.append(
+                torch.zeros(1, dtype=torch.float32, device=self._device)
+            )
+        return self._loss_scales[idx]
+
+    def backward_per_loss(self, losses, optimizer, params):
+        """Backward every loss under its own scale, unscaling each
+        contribution in place (fused HIP kernel) before the next loss."""
+        state = self._per_optimizer_states[id(optimizer)]
+        found = torch.zeros(1, dtype=torch.float32, device=self._device)
+        for idx, loss in enumerate(losses):
+            sc = self._loss_state(idx)
+            stash = []
+            for p in params:
+                stash.append(p.grad)
+                p.grad = None
+            scaled = loss * sc.to(loss.device).to(loss.dtype)
+            scaled.backward(retain_graph=(idx == 0))
+            fi = self._loss_found[idx]
+            fi.zero_()
+            by_dev = defaultdict(list)
+            for p in params:
+                if p.grad is not None:
+                    by_dev[(p.grad.device, p.grad.dtype)].append(p.grad)
+            inv = sc.reciprocal()
+            for (dev, _dt), gs in by_dev.items():
+                ops.multi_tensor_unscale_(gs, inv.to(dev), fi)
+            found = torch.maximum(found, fi.to(found.device))
+            for p, s in zip(params, stash):
+                if s is not None:
+                    p.grad = s if p.grad is None else p.grad.add_(s)
+        state["found_inf"] = found
+        state["unscaled"] = True  # grads are already true units
+
+    def unscale_(self, optimizer):
+        state = self._per_optimizer_states[id(optimizer)]
+        if state["unscaled"]:
+            return  # per-loss backward already unscaled in place
+        super().unscale_(optimizer)
+
+    def update(self, new_scale=None):
+        if new_scale is not None:
+            return super().update(new_scale)
+        self._lazy_init()
+        for sc, tr, fi in zip(self._loss_scales, self._loss_trackers,
+                              self._loss_found):
+            ops.amp_update_scale_(sc, tr, fi, self._growth_factor,
+                                  self._backoff_factor, self._growth_interval)
+        self._per_optimizer_states.clear()
+
+    def get_scale(self) -> float:
+        if not self._enabled:
+            return 1.0
+        if self._loss_scales:
+            return self._loss_scales[0].item()
+        return super().get_scale()
+
+    def state_dict(self):
+        sd = super().state_dict()
+        sd["per_loss_scales"] = [s.item() for s in self._loss_scales]
+        sd["per_loss_trackers"] = [int(t.item()) for t in self._loss_trackers]
+        return sd
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        for i, (s, t) in enumerate(zip(state_dict.get("per_loss_scales", []),
+                                       state_dict.get("per_loss_trackers", []))):
+            self._loss_state(i)
+            self._loss_scales[i].fill_(float(s))
+            self._loss_trackers[i].fill_(int(t))

@@ -253,6 +253,26 @@ class StokeDDPModule(torch.nn.Module):
         if self._callback_queued or any(b.launched for b in self._buckets):
             self._finalize_backward()
 
+    def sync_existing_grads(self):
+        """All-reduce whatever is in ``p.grad`` right now, one blocking
+        pass (used after per-loss-scaled backwards, where the in-backward
+        hooks were suppressed via ``no_sync``)."""
+        if self._pg.world_size == 1:
+            return
+        prev_unused = self._find_unused
+        prev_sync = self.require_backward_grad_sync
+        self._find_unused = True  # launch every bucket, grads are in place
+        self.require_backward_grad_sync = True
+        self._callback_queued = False
+        for b in self._buckets:
+            b.launched = False
+            b.ready = 0
+        try:
+            self._finalize_backward()
+        finally:
+            self._find_unused = prev_unused
+            self.require_backward_grad_sync = prev_sync
+
     @contextmanager
     def no_sync(self):
         """Suppress gradient synchronization (gradient-accumulation context)."""

@@ -24,7 +24,7 @@ from typing import List, Optional, Tuple, Union
 
 import torch
 
-from stoke.amp import StokeGradScaler
+from stoke.amp import StokeGradScaler, StokePerLossScaler
 from stoke.comm import StokeProcessGroup
 from stoke.configs import ClipGradConfig, ClipGradNormConfig
 from stoke.ddp import StokeDDPModule
@@ -120,7 +120,20 @@ class StokeRunner:
                 )
             else:
                 amp = self._status.amp_config
-                self._scaler = StokeGradScaler(
+                # apex parity: scaler_per_loss gives every loss index its
+                # own dynamic scale (reference fp16.py:545-579)
+                per_loss = bool(
+                    getattr(self._status.apex_config, "scaler_per_loss", False)
+                    and str(self._status.fp16 or "").startswith("apex")
+                )
+                cls = StokePerLossScaler if per_loss else StokeGradScaler
+                if per_loss and self._shard in ("sddp", "fsdp"):
+                    raise NotImplementedError(
+                        "Stoke -- scaler_per_loss is not supported with "
+                        "sharded-gradient modes (sddp/fsdp); use the global "
+                        "scaler or unsharded DDP"
+                    )
+                self._scaler = cls(
                     init_scale=amp.init_scale,
                     growth_factor=amp.growth_factor,
                     backoff_factor=amp.backoff_factor,
@@ -240,6 +253,14 @@ class StokeRunner:
         # stage-0/1 gradient path)
         dcfg = s.ddp_config if not s.is_distributed_deepspeed else None
         hv = s.horovod_config if s.is_distributed_horovod else None
+        if hv is not None and getattr(hv.op, "value", hv.op) == "Adasum":
+            # The reference delegates Adasum to horovod's adaptive-summation
+            # kernels (distributed.py:1416-1428); mapping it silently onto a
+            # plain SUM would change optimization semantics, so refuse.
+            raise NotImplementedError(
+                "Stoke -- HorovodConfig op 'Adasum' (adaptive summation) is "
+                "not implemented by the RCCL engine; use 'Average' or 'Sum'"
+            )
         wrapped = StokeDDPModule(
             model,
             pg=self._pg,
@@ -337,6 +358,21 @@ class StokeRunner:
     # ------------------------------------------------------- backward / step
     def backward_call(self, loss, model, optimizer):
         self._grads_ready = False
+        if (isinstance(loss, (list, tuple))
+                and isinstance(self._scaler, StokePerLossScaler)):
+            # per-loss scales: each backward is stash-isolated + unscaled;
+            # gradient sync is deferred to one pass after the last loss
+            # (the DDP hooks must not reduce scaled partial grads)
+            params = [p for g in optimizer.param_groups for p in g["params"]]
+            engine = self._engine
+            if engine is not None and hasattr(engine, "no_sync"):
+                with engine.no_sync():
+                    self._scaler.backward_per_loss(loss, optimizer, params)
+                if hasattr(engine, "sync_existing_grads"):
+                    engine.sync_existing_grads()
+            else:
+                self._scaler.backward_per_loss(loss, optimizer, params)
+            return
         if self._scaler is not None:
             scaled = self._scaler.scale(loss)
         else:

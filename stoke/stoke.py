@@ -434,13 +434,24 @@ class Stoke:
             self._optimizer_steps += 1
             if (
                 self._flops_profiler is not None
+                and self._optimizer_steps == self._flops_cfg.profile_step - 1
+            ):
+                # zero counters so the profiled step counts exactly one
+                # optimizer step's forwards
+                self._flops_profiler.reset_flops()
+            if (
+                self._flops_profiler is not None
                 and self._optimizer_steps >= self._flops_cfg.profile_step
             ):
-                self._flops_profiler.print_model_profile(
-                    top_modules=self._flops_cfg.top_modules,
-                    detailed=self._flops_cfg.detailed,
-                    output_file=self._flops_cfg.output_file,
-                )
+                # rank-0 only: every rank printing (and racing on
+                # output_file) was an ADVICE.md round-1 finding
+                rank = self._runner.rank
+                if rank in ("cpu", "gpu", 0):
+                    self._flops_profiler.print_model_profile(
+                        top_modules=self._flops_cfg.top_modules,
+                        detailed=self._flops_cfg.detailed,
+                        output_file=self._flops_cfg.output_file,
+                    )
                 self._flops_profiler.stop_profile()
                 self._flops_profiler = None
         elif self.is_deepspeed:

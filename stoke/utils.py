@@ -154,6 +154,14 @@ class FlopsProfiler:
                 self.params[name] = sum(p.numel() for p in mod.parameters())
                 self._handles.append(mod.register_forward_hook(make_hook(name)))
 
+    def reset_flops(self):
+        """Zero the FLOP counters while keeping hooks attached — called at
+        the step boundary before the profiled step so the printed numbers
+        cover exactly ONE optimizer step (accumulating
+        from construction inflated the profile by the number of forwards
+        under profile_step > 1 / gradient accumulation)."""
+        self.flops.clear()
+
     def stop_profile(self):
         for h in self._handles:
             h.remove()
