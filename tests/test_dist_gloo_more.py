@@ -297,3 +297,48 @@ def test_sddp_fp16_reduce_prediv_no_overflow():
     torch.multiprocessing.spawn(
         _sddp_fp16_worker, args=(2, free_port()), nprocs=2, join=True
     )
+
+
+# ------------------------------------------- per-loss scaler + DDP engine
+def _per_loss_ddp_worker(rank, world, port):
+    import torch.nn as nn
+    from tests.conftest import init_gloo
+    from stoke.amp import StokePerLossScaler
+    from stoke.ddp import StokeDDPModule
+
+    pg = init_gloo(rank, world, port)
+    torch.manual_seed(0)
+    model = nn.Linear(8, 4)
+    ddp = StokeDDPModule(model, pg=pg, bucket_cap_mb=1)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    sc = StokePerLossScaler(init_scale=8.0, device="cpu")
+    params = list(model.parameters())
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    y = torch.randint(0, 4, (4,))
+    with ddp.no_sync():
+        losses = [nn.CrossEntropyLoss()(ddp(x), y),
+                  0.5 * nn.CrossEntropyLoss()(ddp(x), y)]
+        sc.backward_per_loss(losses, opt, params)
+    ddp.sync_existing_grads()
+    # grads equal across ranks (all-reduced true-unit grads)
+    for p in params:
+        flat = p.grad.reshape(-1).clone()
+        torch.distributed.broadcast(flat, src=0)
+        assert torch.allclose(flat, p.grad.reshape(-1), atol=1e-6)
+    sc.step(opt)
+    sc.update()
+    # params stay equal across ranks
+    for p in params:
+        flat = p.detach().reshape(-1).clone()
+        torch.distributed.broadcast(flat, src=0)
+        assert torch.allclose(flat, p.detach().reshape(-1), atol=1e-6)
+    torch.distributed.destroy_process_group()
+
+
+def test_per_loss_scaler_with_ddp_sync():
+    from tests.conftest import free_port
+
+    torch.multiprocessing.spawn(
+        _per_loss_ddp_worker, args=(2, free_port()), nprocs=2, join=True
+    )

@@ -95,3 +95,94 @@ def test_sharded_scaler_syncs_found_inf():
     torch.multiprocessing.spawn(
         _sharded_scaler_worker, args=(2, free_port()), nprocs=2, join=True
     )
+
+
+# --------------------------------------------------- per-loss scaler (CPU)
+def test_per_loss_scaler_independent_backoff():
+    import torch
+    from stoke.amp import StokePerLossScaler
+
+    p = torch.nn.Parameter(torch.ones(8))
+    opt = torch.optim.SGD([p], lr=0.1)
+    sc = StokePerLossScaler(init_scale=2.0**8, device="cpu")
+    x = torch.ones(8)
+
+    def losses():
+        a = (p * x).sum() * 1.0
+        b = (p * x).sum() * float("inf")  # loss 1 overflows every step
+        return [a, b]
+
+    sc.backward_per_loss(losses(), opt, [p])
+    before = p.detach().clone()
+    sc.step(opt)
+    sc.update()
+    # inf in loss-1's contribution -> step skipped
+    assert torch.equal(p.detach(), before)
+    # per-loss: scale 0 untouched (growth pending), scale 1 backed off
+    assert sc._loss_scales[0].item() == 2.0**8
+    assert sc._loss_scales[1].item() == 2.0**7
+    # finite-only losses step normally and grads are true units
+    opt.zero_grad()
+    a = (p * x).sum()
+    sc.backward_per_loss([a, a * 2.0], opt, [p])
+    assert torch.allclose(p.grad, torch.full((8,), 3.0), atol=1e-5)
+    sc.step(opt)
+    sc.update()
+    assert not torch.equal(p.detach(), before)
+
+
+def test_per_loss_scaler_state_roundtrip():
+    import torch
+    from stoke.amp import StokePerLossScaler
+
+    sc = StokePerLossScaler(init_scale=4.0, device="cpu")
+    sc._loss_state(1)
+    sc._loss_scales[1].fill_(16.0)
+    sd = sc.state_dict()
+    sc2 = StokePerLossScaler(device="cpu")
+    sc2.load_state_dict(sd)
+    assert sc2._loss_scales[1].item() == 16.0
+
+
+def test_horovod_adasum_raises():
+    import pytest
+    import torch
+    from stoke import Stoke, StokeOptimizer, HorovodConfig
+
+    # status requires CUDA for distributed; shim probes like the gloo tests
+    torch.cuda.is_available = lambda: True
+    torch.distributed.is_nccl_available = lambda: True
+    try:
+        import os
+
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29631",
+                          RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+        torch.cuda.set_device = lambda *a, **k: None
+        torch.nn.Module.cuda = lambda self, *a, **k: self
+        from stoke import DDPConfig
+
+        model = torch.nn.Linear(4, 2)
+        with pytest.raises(NotImplementedError, match="Adasum"):
+            Stoke(
+                model=model,
+                optimizer=StokeOptimizer(optimizer=torch.optim.SGD,
+                                         optimizer_kwargs={"lr": 0.1}),
+                loss=torch.nn.MSELoss(),
+                batch_size_per_device=2,
+                gpu=True,
+                distributed="horovod",
+                configs=[DDPConfig(local_rank=0, backend="gloo"),
+                         HorovodConfig(op="Adasum")],
+                verbose=False,
+            )
+    finally:
+        if torch.distributed.is_initialized():
+            torch.distributed.destroy_process_group()
+
+
+def test_git_version():
+    import stoke
+    from stoke._version import get_versions
+
+    v = get_versions()
+    assert v["version"] and stoke.__version__ == v["version"]
