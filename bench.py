@@ -111,7 +111,12 @@ def main():
         # Exhaustive MIOpen find maximizes steady-state conv speed but costs
         # ~4-5 min of warmup on a box with a cold find-db (amortized: later
         # runs on the same box reuse it).  STOKE_FAST_FIND=1 trades a few %
-        # of conv throughput for heuristic-mode startup.
+        # of conv throughput for heuristic-mode startup.  Multi-rank runs
+        # default to FAST (even when launched by an external torchrun):
+        # N ranks of exhaustive find on a cold box would eat the whole
+        # launcher timeout, and FAST still consults a warm find-db first.
+        if distributed and not os.environ.get("STOKE_FULL_FIND"):
+            os.environ.setdefault("STOKE_FAST_FIND", "1")
         if os.environ.get("STOKE_FAST_FIND"):
             os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
         else:
