@@ -107,8 +107,20 @@ class FP8Linear(nn.Linear):
         return super().forward(x)
 
 
-def convert_linears_to_fp8(model: nn.Module, min_features: int = 1024) -> int:
-    """Swap every big-enough nn.Linear for FP8Linear in place; returns count."""
+def convert_linears_to_fp8(model: nn.Module, min_features: int = 1024,
+                           delayed: bool = True) -> int:
+    """Swap every big-enough nn.Linear for an fp8 linear in place.
+
+    ``delayed=True`` (default) uses :class:`FP8LinearDelayed` — scales come
+    from the previous step's amax (no extra reduction pass; the next amax is
+    a byproduct of the fused cast kernel) and the backward's column-major
+    operands come from the dual-layout quantizer.  ``delayed=False`` keeps
+    the per-call-amax v1 path.  Returns the number of swapped layers.
+    """
+    if delayed:
+        from stoke.nn.fp8_delayed import FP8LinearDelayed as cls
+    else:
+        cls = FP8Linear
     n = 0
     for mod in model.modules():
         for name, child in list(mod.named_children()):
@@ -118,8 +130,8 @@ def convert_linears_to_fp8(model: nn.Module, min_features: int = 1024) -> int:
                 and child.in_features % 16 == 0
                 and child.out_features % 16 == 0
             ):
-                fp8 = FP8Linear(child.in_features, child.out_features,
-                                bias=child.bias is not None)
+                fp8 = cls(child.in_features, child.out_features,
+                          bias=child.bias is not None)
                 fp8.weight = child.weight
                 if child.bias is not None:
                     fp8.bias = child.bias
