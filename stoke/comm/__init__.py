@@ -179,18 +179,22 @@ class StokeProcessGroup:
                 t.copy_(flat[offset : offset + n].view_as(t))
                 offset += n
 
-    def sync_loss(self, loss: torch.Tensor, device=None) -> float:
-        """Mean-reduce a scalar loss across ranks and return a python float.
+    def sync_loss(self, loss: torch.Tensor, device=None):
+        """Mean-reduce a scalar loss across ranks, ASYNCHRONOUSLY.
 
-        Unlike the reference (D2H sync + barrier + all_reduce every microbatch,
-        ``distributed.py:619-646``) this is a single fused all-reduce of one
-        scalar followed by one D2H read — no barrier.
+        Returns a float-like :class:`stoke.utils.LazyLoss`: the all-reduce
+        is launched async (no barrier, unlike the reference at
+        ``distributed.py:619-646``) and the one D2H read happens only when
+        a real float is needed — per-microbatch loss tracking no longer
+        stalls the host.
         """
+        from stoke.utils import LazyLoss
+
         t = loss.detach().clone().float()
         if self.world_size > 1:
-            dist.all_reduce(t)
-            t /= self.world_size
-        return t.item()
+            work = dist.all_reduce(t, async_op=True)
+            return LazyLoss(t, work=work, div=self.world_size)
+        return LazyLoss(t)
 
     def clean(self):
         if dist.is_initialized():

@@ -564,7 +564,15 @@ class Stoke:
     # ------------------------------------------------------------- properties
     @property
     def step_loss(self):
-        return self._last_step_loss
+        return self._coerce_loss(self._last_step_loss)
+
+    @staticmethod
+    def _coerce_loss(val):
+        """Public properties expose plain floats; internal tracking may hold
+        lazy (async-synced) values."""
+        if isinstance(val, (list, tuple)):
+            return type(val)(float(v) for v in val)
+        return float(val)
 
     @property
     def model_access(self):
@@ -709,4 +717,4 @@ class Stoke:
 
     @property
     def ema_loss(self):
-        return self._rolling_mean_loss
+        return self._coerce_loss(self._rolling_mean_loss)

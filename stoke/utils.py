@@ -193,3 +193,99 @@ class FlopsProfiler:
         else:
             unrolled_print(lines)
         return text
+
+
+class LazyLoss:
+    """Float-like handle on an in-flight loss all-reduce.
+
+    ``sync_loss`` launches the scalar all-reduce asynchronously and returns
+    one of these instead of blocking on ``.item()`` every micro-batch
+    (the reference D2H-synced + barriered per
+    micro-batch at ``distributed.py:619-646``).  Arithmetic (the facade's
+    agg/EMA bookkeeping) stays device-side and lazy; the single host sync
+    happens only when a float is actually needed (print helpers,
+    ``float()``, ``format``).
+    """
+
+    __slots__ = ("_t", "_work", "_div")
+
+    def __init__(self, t, work=None, div=1.0):
+        self._t = t
+        self._work = work
+        self._div = float(div)
+
+    # -------------------------------------------------------------- plumbing
+    def _tensor(self):
+        if self._work is not None:
+            self._work.wait()
+            self._work = None
+        if self._div != 1.0:
+            self._t = self._t / self._div
+            self._div = 1.0
+        return self._t
+
+    @staticmethod
+    def _raw(x):
+        return x._tensor() if isinstance(x, LazyLoss) else x
+
+    def item(self) -> float:
+        return float(self._tensor().item())
+
+    # ------------------------------------------------------------ float-like
+    def __float__(self):
+        return self.item()
+
+    def __format__(self, spec):
+        return format(self.item(), spec)
+
+    def __repr__(self):
+        return repr(self.item())
+
+    def __add__(self, other):
+        return LazyLoss(self._tensor() + self._raw(other))
+
+    __radd__ = __add__
+
+    def __sub__(self, other):
+        return LazyLoss(self._tensor() - self._raw(other))
+
+    def __rsub__(self, other):
+        return LazyLoss(self._raw(other) - self._tensor())
+
+    def __mul__(self, other):
+        return LazyLoss(self._tensor() * self._raw(other))
+
+    __rmul__ = __mul__
+
+    def __truediv__(self, other):
+        return LazyLoss(self._tensor() / self._raw(other))
+
+    def __rtruediv__(self, other):
+        return LazyLoss(self._raw(other) / self._tensor())
+
+    def __neg__(self):
+        return LazyLoss(-self._tensor())
+
+    def __abs__(self):
+        return LazyLoss(self._tensor().abs())
+
+    def __round__(self, ndigits=None):
+        return round(self.item(), ndigits)
+
+    def __lt__(self, other):
+        return self.item() < float(self._raw(other))
+
+    def __le__(self, other):
+        return self.item() <= float(self._raw(other))
+
+    def __gt__(self, other):
+        return self.item() > float(self._raw(other))
+
+    def __ge__(self, other):
+        return self.item() >= float(self._raw(other))
+
+    def __eq__(self, other):
+        try:
+            return self.item() == float(self._raw(other))
+        except (TypeError, ValueError):
+            return NotImplemented
