@@ -422,51 +422,74 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
 
   const float scale = rsqrtf((float)DH);
   const int tid = threadIdx.x;
-  const bool is_rm = tid < 256;
-  const int ka = tid & 255;
-  constexpr int RPIECE = FB_QB * KSLOT / 256;
-  const int v_dchunk = ka >> 4;
-  const int v_kvq = ka & 15;
-  const bool tr_active = !is_rm && v_dchunk < DH / 8;
 
-  const int kv_end = causal ? min(qbase + 256, S) : S;
-  for (int kb = 0; kb < kv_end; kb += FB_QB) {
-    __syncthreads();
-    if (is_rm) {
-      #pragma unroll
-      for (int i = 0; i < RPIECE; ++i) {
-        const int p = ka * RPIECE + i;
-        const int row = p / KSLOT;
-        const int slot = p % KSLOT;
-        const int grow = min(kb + row, S - 1);
-        const int sslot = slot ^ (row & (KSLOT - 1));
-        *reinterpret_cast<bf16x8v*>(&krm[row * DH + sslot * 8]) =
-            *reinterpret_cast<const bf16x8v*>(
-                kp + (long)grow * kss + slot * 8);
-        *reinterpret_cast<bf16x8v*>(&vrm[row * DH + sslot * 8]) =
-            *reinterpret_cast<const bf16x8v*>(
-                vp + (long)grow * vss + slot * 8);
-      }
-    } else if (tr_active) {
-      union { bf16x8v v8[4]; short s[4][8]; } u;
+  // ---- tile pipeline (cf. fa_bwd_dkv_v1): K/V row-major pieces for tile
+  // t+1 prefetched to registers during tile t's compute; the transposed K
+  // image is built from the row-major LDS image.
+  constexpr int NT = 512;
+  constexpr int NPREF = 2 * FB_QB * KSLOT / NT;
+  bf16x8v pref[NPREF];
+
+  auto load_rm = [&](int kb) {
+    #pragma unroll
+    for (int i = 0; i < NPREF; ++i) {
+      const int p = tid + i * NT;
+      const int img = p >= FB_QB * KSLOT;
+      const int pp = p - img * FB_QB * KSLOT;
+      const int row = pp / KSLOT;
+      const int slot = pp % KSLOT;
+      const int grow = min(kb + row, S - 1);
+      const bf16* src = img == 0 ? kp : vp;
+      const long rs = img == 0 ? kss : vss;
+      pref[i] = *reinterpret_cast<const bf16x8v*>(
+          src + (long)grow * rs + slot * 8);
+    }
+  };
+
+  auto write_rm = [&]() {
+    #pragma unroll
+    for (int i = 0; i < NPREF; ++i) {
+      const int p = tid + i * NT;
+      const int img = p >= FB_QB * KSLOT;
+      const int pp = p - img * FB_QB * KSLOT;
+      const int row = pp / KSLOT;
+      const int slot = pp % KSLOT;
+      const int sslot = slot ^ (row & (KSLOT - 1));
+      bf16* dst = img == 0 ? krm : vrm;
+      *reinterpret_cast<bf16x8v*>(&dst[row * DH + sslot * 8]) = pref[i];
+    }
+  };
+
+  auto build_tr = [&]() {
+    for (int a = tid; a < (DH / 8) * 16; a += NT) {
+      const int dchunk = a >> 4;
+      const int rq = a & 15;
+      union { bf16x8v v8[4]; short sh[4][8]; } u;
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int grow = min(kb + v_kvq * 4 + i, S - 1);
+        const int row = rq * 4 + i;
+        const int sslot = dchunk ^ (row & (KSLOT - 1));
         u.v8[i] = *reinterpret_cast<const bf16x8v*>(
-            kp + (long)grow * kss + v_dchunk * 8);
+            &krm[row * DH + sslot * 8]);
       }
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         union { short s4[4]; unsigned long long d; } pack;
         #pragma unroll
-        for (int i = 0; i < 4; ++i) pack.s4[i] = u.s[i][j];
+        for (int i = 0; i < 4; ++i) pack.s4[i] = u.sh[i][j];
         *reinterpret_cast<unsigned long long*>(
-            &ktr[(v_dchunk * 8 + j) * (FB_QB + FB_PAD) + v_kvq * 4]) =
-            pack.d;
+            &ktr[(dchunk * 8 + j) * (FB_QB + FB_PAD) + rq * 4]) = pack.d;
       }
     }
-    __syncthreads();
+  };
 
+  const int kv_end = causal ? min(qbase + 256, S) : S;
+  load_rm(0);
+  write_rm();
+  __syncthreads();
+  build_tr();
+  __syncthreads();
+  for (int kb = 0; kb < kv_end; kb += FB_QB) {
     #pragma unroll
     for (int ct = 0; ct < 2; ++ct) {
       f32x16v sacc, dpacc;
@@ -475,6 +498,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
         sacc[r] = 0.f;
         dpacc[r] = 0.f;
       }
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int ch = 0; ch < DCH; ++ch) {
         const int row = ct * 32 + lq;
@@ -488,6 +512,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
         dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ch], dpacc,
                                                         0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
       // C[kv, q]: lane = q column; elementwise uses the lane's own L/delta
       const int qg = row0 + lq;
       #pragma unroll
@@ -500,6 +525,9 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
       }
       bf16x8v dsb[2];
       fb_repack16(dpacc, h2, dsb);
+      if (ct == 0 && kb + FB_QB < kv_end)
+        load_rm(kb + FB_QB);  // overlap next tile's HBM latency
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         #pragma unroll
@@ -511,6 +539,14 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
               kta, dsb[kc], dqacc[dt], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    if (kb + FB_QB < kv_end) {
+      write_rm();
+      __syncthreads();
+      build_tr();
+      __syncthreads();
     }
   }
 
