@@ -53,12 +53,6 @@ def _maybe_self_launch(args):
 
     if args.gpus <= 1 or "WORLD_SIZE" in os.environ:
         return
-    # 8 ranks of exhaustive MIOpen find can eat the whole driver timeout on
-    # a cold box; FAST mode consults the find-db first (so a prior N=1
-    # exhaustive run still gives full-speed convs) and falls back to the
-    # heuristic instead of a multi-minute search.  STOKE_FULL_FIND=1 opts out.
-    if not os.environ.get("STOKE_FULL_FIND"):
-        os.environ.setdefault("STOKE_FAST_FIND", "1")
     port = str(29400 + os.getpid() % 1000)
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
@@ -108,18 +102,21 @@ def main():
         dtype = "fp32"
         mname = "resnet18-cifar10-shape-cpu"
     elif args.model == "resnet50":
-        # Exhaustive MIOpen find maximizes steady-state conv speed but costs
-        # ~4-5 min of warmup on a box with a cold find-db (amortized: later
-        # runs on the same box reuse it).  STOKE_FAST_FIND=1 trades a few %
-        # of conv throughput for heuristic-mode startup.  Multi-rank runs
-        # default to FAST (even when launched by an external torchrun):
-        # N ranks of exhaustive find on a cold box would eat the whole
-        # launcher timeout, and FAST still consults a warm find-db first.
-        if distributed and not os.environ.get("STOKE_FULL_FIND"):
-            os.environ.setdefault("STOKE_FAST_FIND", "1")
-        if os.environ.get("STOKE_FAST_FIND"):
+        # MIOpen find policy (measured on MI355X):
+        #   exhaustive (cudnn.benchmark=True): fastest steady state
+        #     (8.2k samples/s) but ~4-5 min of one-off find on a cold box
+        #     (the per-box find-db amortizes later runs, including the
+        #     other ranks of a scaling sweep).
+        #   MIOpen default (HYBRID): no multi-minute search; used for
+        #     multi-rank runs so N ranks of cold find cannot eat the
+        #     launcher timeout.  A warm find-db still gives full speed.
+        #   FAST: MEASURED 40x SLOWER on this workload (213 samples/s,
+        #     gpurun_out/call12.log — naive conv fallback); never default.
+        find = os.environ.get(
+            "STOKE_FIND", "default" if distributed else "exhaustive")
+        if find == "fast" or os.environ.get("STOKE_FAST_FIND"):
             os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
-        else:
+        elif find == "exhaustive":
             torch.backends.cudnn.benchmark = True
         model = models.resnet50(num_classes=1000)
         # 384/GPU measured fastest on MI355X (8150 samples/s vs 7765 at 256;

@@ -47,11 +47,11 @@ __device__ __forceinline__ float fb_b2f(const bf16& h) {
   return __bfloat162float(h);
 }
 
+// pack two f32 -> two bf16 in one dword (one v_cvt_pk_bf16_f32 — the
+// scalar cvt+or sequence costs 3-4 VALU ops per dword; guide T12)
 __device__ __forceinline__ unsigned int fb_pk(float lo, float hi) {
-  bf16 a = __float2bfloat16(lo);
-  bf16 b = __float2bfloat16(hi);
-  unsigned int r = (unsigned int)*reinterpret_cast<unsigned short*>(&a);
-  r |= ((unsigned int)*reinterpret_cast<unsigned short*>(&b)) << 16;
+  unsigned int r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
   return r;
 }
 
