@@ -146,6 +146,8 @@ at::Tensor fp8_quant(at::Tensor x, at::Tensor scale, at::Tensor amax_next,
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous() &&
                   x.numel() % 8 == 0,
               "fp8_quant: contiguous bf16, numel %% 8 == 0");
+  TORCH_CHECK(scale.is_cuda() && amax_next.is_cuda(),
+              "fp8_quant: scale/amax must be CUDA tensors");
   auto dt = kind == 0 ? at::ScalarType::Float8_e4m3fn
                       : at::ScalarType::Float8_e5m2;
   auto y = at::empty_like(x, x.options().dtype(dt));
@@ -175,6 +177,10 @@ std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.dim() == 2 &&
                   x.is_contiguous(),
               "fp8_quant_t: contiguous 2-D bf16");
+  TORCH_CHECK(scale.is_cuda() && amax_next.is_cuda() &&
+                  scale.scalar_type() == at::kFloat &&
+                  amax_next.scalar_type() == at::kFloat,
+              "fp8_quant_t: scale/amax must be fp32 CUDA tensors");
   const int M = (int)x.size(0), N = (int)x.size(1);
   auto dt = kind == 0 ? at::ScalarType::Float8_e4m3fn
                       : at::ScalarType::Float8_e5m2;

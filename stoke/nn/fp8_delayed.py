@@ -101,11 +101,15 @@ class FP8LinearDelayed(nn.Linear):
         )
         if not usable:
             return super().forward(x)
-        if self._sx.dtype != torch.float32:
-            # module-level .bfloat16()/.half() converts registered buffers;
-            # the quantizer contract is fp32 scales/amax — restore them
+        if self._sx.dtype != torch.float32 or self._sx.device != x.device:
+            # module-level .bfloat16()/.half() converts registered buffers,
+            # and converting an already-on-GPU model leaves fresh buffers on
+            # the CPU; the quantizer contract is fp32 scales/amax on the
+            # input's device — restore both
             for name in ("_sx", "_sw", "_sdy", "_ax", "_aw", "_ady"):
-                setattr(self, name, getattr(self, name).float())
+                setattr(self, name,
+                        getattr(self, name).to(device=x.device,
+                                               dtype=torch.float32))
         if self._primed:
             self._update_scales()
         shape = x.shape[:-1] + (self.out_features,)
