@@ -46,6 +46,27 @@ __device__ __forceinline__ unsigned char fq_cast(float f) {
   return *reinterpret_cast<unsigned char*>(&q);
 }
 
+
+// Hardware packed converts (4 VALU ops per 8 values instead of 8 software
+// constructor calls): v_cvt_pk_fp8_f32 / v_cvt_pk_bf8_f32.
+template <bool E4M3>
+__device__ __forceinline__ unsigned int fq_pk4(float a, float b, float c,
+                                               float d, float mx) {
+  a = fminf(fmaxf(a, -mx), mx);
+  b = fminf(fmaxf(b, -mx), mx);
+  c = fminf(fmaxf(c, -mx), mx);
+  d = fminf(fmaxf(d, -mx), mx);
+  unsigned int r = 0;
+  if (E4M3) {
+    r = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, r, false);
+    r = __builtin_amdgcn_cvt_pk_fp8_f32(c, d, r, true);
+  } else {
+    r = __builtin_amdgcn_cvt_pk_bf8_f32(a, b, r, false);
+    r = __builtin_amdgcn_cvt_pk_bf8_f32(c, d, r, true);
+  }
+  return r;
+}
+
 // atomic max for non-negative floats via ordered uint bits
 __device__ __forceinline__ void fq_atomic_amax(float* addr, float v) {
   atomicMax(reinterpret_cast<unsigned int*>(addr),
@@ -53,7 +74,7 @@ __device__ __forceinline__ void fq_atomic_amax(float* addr, float v) {
 }
 
 // y8[i] = fp8(x[i] / *scale); *amax_next = max|x| (block-reduced)
-template <typename F8>
+template <bool E4M3>
 __global__ __launch_bounds__(FQ_BLOCK) void fp8_quant_kernel(
     const fq_bf16x8* __restrict__ x, unsigned char* __restrict__ y,
     const float* __restrict__ scale, float* __restrict__ amax_next,
@@ -64,14 +85,17 @@ __global__ __launch_bounds__(FQ_BLOCK) void fp8_quant_kernel(
   for (long i = (long)blockIdx.x * FQ_BLOCK + threadIdx.x; i < n8;
        i += stride) {
     fq_bf16x8 vv = x[i];
-    unsigned char o[8];
+    float f[8];
     #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      float f = fq_b2f(vv.h[e]);
-      am = fmaxf(am, fabsf(f));
-      o[e] = fq_cast<F8>(fminf(fmaxf(f * inv, -fp8_max), fp8_max));
+      const float raw = fq_b2f(vv.h[e]);
+      am = fmaxf(am, fabsf(raw));
+      f[e] = raw * inv;
     }
-    *reinterpret_cast<uint2*>(y + i * 8) = *reinterpret_cast<uint2*>(o);
+    uint2 o;
+    o.x = fq_pk4<E4M3>(f[0], f[1], f[2], f[3], fp8_max);
+    o.y = fq_pk4<E4M3>(f[4], f[5], f[6], f[7], fp8_max);
+    *reinterpret_cast<uint2*>(y + i * 8) = o;
   }
   // block-reduce the amax, one atomic per block
   __shared__ float smem[FQ_BLOCK / 64];
@@ -87,42 +111,84 @@ __global__ __launch_bounds__(FQ_BLOCK) void fp8_quant_kernel(
   }
 }
 
-// Row-major [M,N] -> y8 row-major AND yt8 [N,M]; 32x32 LDS transpose tiles.
-// Block = 256 threads = 8 rows of 32 lanes per pass; grid 2-D over tiles.
-template <typename F8>
+// Row-major [M,N] -> y8 row-major AND yt8 [N,M]; 64x64 LDS transpose tiles.
+// Vectorized: 16-B bf16 loads, packed fp8 converts (4 VALU / 8 values),
+// 8-B stores on BOTH layouts (the scalar 1-2-B version measured 65% of the
+// whole fp8 step — gpurun_out/prof2/fp8_kernel_stats.csv).
+template <bool E4M3>
 __global__ __launch_bounds__(FQ_BLOCK) void fp8_quant_t_kernel(
     const bf16* __restrict__ x, unsigned char* __restrict__ y,
     unsigned char* __restrict__ yt, const float* __restrict__ scale,
     float* __restrict__ amax_next, int M, int N, float fp8_max) {
-  __shared__ unsigned char tile[32][33];  // +1 pad: bank-conflict-free
+  constexpr int TP = 72;                 // padded LDS tile row (bytes)
+  __shared__ unsigned char tile[64 * TP];
   const float inv = 1.f / *scale;
-  const int tx = threadIdx.x & 31;   // col within tile
-  const int ty = threadIdx.x >> 5;   // row group (8 rows/pass, 4 passes)
-  const int col0 = blockIdx.x * 32;
-  const int row0 = blockIdx.y * 32;
+  const int col0 = blockIdx.x * 64;
+  const int row0 = blockIdx.y * 64;
   float am = 0.f;
+  // load+convert phase: 2 passes x (row = t/8, 8-col chunk = t%8)
   #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int r = row0 + p * 8 + ty;
-    const int c = col0 + tx;
-    unsigned char q = 0;
-    if (r < M && c < N) {
-      float f = fq_b2f(*reinterpret_cast<const unsigned short*>(
-          &x[(long)r * N + c]));
-      am = fmaxf(am, fabsf(f));
-      q = fq_cast<F8>(fminf(fmaxf(f * inv, -fp8_max), fp8_max));
-      y[(long)r * N + c] = q;
+  for (int p = 0; p < 2; ++p) {
+    const int a = (int)threadIdx.x + p * FQ_BLOCK;
+    const int lr = a >> 3;               // tile row 0..63
+    const int lc = (a & 7) * 8;          // tile col chunk
+    const int r = row0 + lr;
+    const int c = col0 + lc;
+    uint2 o = {0u, 0u};
+    if (r < M && c + 7 < N) {
+      fq_bf16x8 vv;
+      vv.u4 = *reinterpret_cast<const uint4*>(&x[(long)r * N + c]);
+      float f[8];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float raw = fq_b2f(vv.h[e]);
+        am = fmaxf(am, fabsf(raw));
+        f[e] = raw * inv;
+      }
+      o.x = fq_pk4<E4M3>(f[0], f[1], f[2], f[3], fp8_max);
+      o.y = fq_pk4<E4M3>(f[4], f[5], f[6], f[7], fp8_max);
+      *reinterpret_cast<uint2*>(y + (long)r * N + c) = o;
+    } else if (r < M) {
+      // ragged tail: scalar path
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        unsigned char q8 = 0;
+        if (c + e < N) {
+          const float raw = fq_b2f(*reinterpret_cast<const unsigned short*>(
+              &x[(long)r * N + c + e]));
+          am = fmaxf(am, fabsf(raw));
+          const unsigned int pk =
+              fq_pk4<E4M3>(raw * inv, 0.f, 0.f, 0.f, fp8_max);
+          q8 = (unsigned char)(pk & 0xff);
+          y[(long)r * N + c + e] = q8;
+        }
+        reinterpret_cast<unsigned char*>(&o)[e] = q8;
+      }
     }
-    tile[p * 8 + ty][tx] = q;
+    *reinterpret_cast<uint2*>(&tile[lr * TP + lc]) = o;
   }
   __syncthreads();
+  // transpose phase: thread gathers 8 rows of one column (scalar LDS byte
+  // reads) and emits ONE 8-B store per pass to the [N,M] layout
   #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int tr = p * 8 + ty;           // row within the TRANSPOSED tile
-    const int c = col0 + tr;             // original col -> yt row
-    const int r = row0 + tx;             // original row -> yt col
-    if (c < N && r < M)
-      yt[(long)c * M + r] = tile[tx][tr];
+  for (int p = 0; p < 2; ++p) {
+    const int a = (int)threadIdx.x + p * FQ_BLOCK;
+    const int lc = a >> 3;               // source col 0..63 -> yt row
+    const int lr = (a & 7) * 8;          // source row chunk -> yt col chunk
+    const int c = col0 + lc;
+    const int r = row0 + lr;
+    if (c < N && r < M) {
+      unsigned char o[8];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = tile[(lr + j) * TP + lc];
+      if (r + 7 < M) {
+        *reinterpret_cast<uint2*>(yt + (long)c * M + r) =
+            *reinterpret_cast<uint2*>(o);
+      } else {
+        for (int j = 0; j < 8 && r + j < M; ++j)
+          yt[(long)c * M + r + j] = o[j];
+      }
+    }
   }
   // amax reduction (same pattern as fp8_quant_kernel)
   __shared__ float smem[FQ_BLOCK / 64];
@@ -156,14 +222,14 @@ at::Tensor fp8_quant(at::Tensor x, at::Tensor scale, at::Tensor amax_next,
   int grid = (int)std::min<long>((n8 + FQ_BLOCK - 1) / FQ_BLOCK, 4096);
   const float mx = kind == 0 ? 448.f : 57344.f;
   if (kind == 0)
-    hipLaunchKernelGGL((fp8_quant_kernel<__hip_fp8_e4m3>), dim3(grid),
+    hipLaunchKernelGGL((fp8_quant_kernel<true>), dim3(grid),
                        dim3(FQ_BLOCK), 0, stream,
                        (const fq_bf16x8*)x.data_ptr(),
                        (unsigned char*)y.data_ptr(),
                        scale.data_ptr<float>(), amax_next.data_ptr<float>(),
                        n8, mx);
   else
-    hipLaunchKernelGGL((fp8_quant_kernel<__hip_fp8_e5m2>), dim3(grid),
+    hipLaunchKernelGGL((fp8_quant_kernel<false>), dim3(grid),
                        dim3(FQ_BLOCK), 0, stream,
                        (const fq_bf16x8*)x.data_ptr(),
                        (unsigned char*)y.data_ptr(),
@@ -187,17 +253,17 @@ std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
   auto y = at::empty({M, N}, x.options().dtype(dt));
   auto yt = at::empty({N, M}, x.options().dtype(dt));
   auto stream = at::hip::getCurrentHIPStream().stream();
-  dim3 grid((N + 31) / 32, (M + 31) / 32);
+  dim3 grid((N + 63) / 64, (M + 63) / 64);
   const float mx = kind == 0 ? 448.f : 57344.f;
   if (kind == 0)
-    hipLaunchKernelGGL((fp8_quant_t_kernel<__hip_fp8_e4m3>), grid,
+    hipLaunchKernelGGL((fp8_quant_t_kernel<true>), grid,
                        dim3(FQ_BLOCK), 0, stream, (const bf16*)x.data_ptr(),
                        (unsigned char*)y.data_ptr(),
                        (unsigned char*)yt.data_ptr(),
                        scale.data_ptr<float>(), amax_next.data_ptr<float>(),
                        M, N, mx);
   else
-    hipLaunchKernelGGL((fp8_quant_t_kernel<__hip_fp8_e5m2>), grid,
+    hipLaunchKernelGGL((fp8_quant_t_kernel<false>), grid,
                        dim3(FQ_BLOCK), 0, stream, (const bf16*)x.data_ptr(),
                        (unsigned char*)y.data_ptr(),
                        (unsigned char*)yt.data_ptr(),
