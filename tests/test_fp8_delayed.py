@@ -1,24 +1,15 @@
 # -*- coding: utf-8 -*-
-"""fp8 fused-quantize kernels — ROUND-2 WORK IN PROGRESS.
+"""fp8 fused-quantize kernels + delayed-scaling linear — hardware-validated.
 
-Gated behind STOKE_FP8V2_TEST=1 (plus the gpu marker); the active fp8 path
-(stoke/nn/fp8.py) is unchanged.  Round 2: validate these, then rebuild
-FP8Linear on delayed scaling (scale from the previous step's amax, updated
-as a byproduct of these kernels; backward uses the pre-transposed copies
-so no `.t().contiguous()` remains)."""
-
-import os
+The v2 vectorized quantizers (HW packed fp8 converts, dual-layout output)
+are the default `convert_linears_to_fp8` path and took Llama-3-8B fp8 from
+a 0.81x regression to a 1.28x win over bf16 (NOTES.md), so this file runs
+in the plain `pytest -m gpu` tier."""
 
 import pytest
 import torch
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(
-        not os.environ.get("STOKE_FP8V2_TEST"),
-        reason="round-2 WIP: set STOKE_FP8V2_TEST=1 to run",
-    ),
-]
+pytestmark = [pytest.mark.gpu]
 
 
 def _ext():
