@@ -369,15 +369,20 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
         }
       }
     } else {
-      float* dkp = dkf + base;
-      float* dvp = dvf + base;
+      // per-head fp32 partials at [gh][B,HKV,S,D]; summed host-side
+      const long pbase = (long)gh * (long)gridDim.y * S * DH +
+                         base;  // gridDim.y == B*HKV
+      float* dkp = dkf + pbase;
+      float* dvp = dvf + pbase;
       #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         #pragma unroll
-        for (int r = 0; r < 16; ++r) {
+        for (int r = 0; r < 16; r += 2) {
           const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
-          atomicAdd(dkp + d, dkacc[dt][r]);
-          atomicAdd(dvp + d, dvacc[dt][r]);
+          *reinterpret_cast<float2*>(dkp + d) =
+              make_float2(dkacc[dt][r], dkacc[dt][r + 1]);
+          *reinterpret_cast<float2*>(dvp + d) =
+              make_float2(dvacc[dt][r], dvacc[dt][r + 1]);
         }
       }
     }
@@ -621,8 +626,9 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const int G = H / HKV;
   at::Tensor dkf, dvf;
   if (G > 1) {
-    dkf = at::zeros({B, HKV, S, DH}, q.options().dtype(at::kFloat));
-    dvf = at::zeros({B, HKV, S, DH}, q.options().dtype(at::kFloat));
+    // per-head partials, plain stores (atomics measured 12% slower overall)
+    dkf = at::empty({G, B, HKV, S, DH}, q.options().dtype(at::kFloat));
+    dvf = at::empty({G, B, HKV, S, DH}, q.options().dtype(at::kFloat));
   }
   const auto LKV = [&](auto kern, dim3 g, dim3 blk) {
     g.z = G;
@@ -658,8 +664,8 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
     LQ(fa_bwd_dq_v1<64>);
   }
   if (G > 1) {
-    dk.copy_(dkf);
-    dv.copy_(dvf);
+    dk.copy_(dkf.sum(0));
+    dv.copy_(dvf.sum(0));
   }
   return {dq, dk, dv};
 }
