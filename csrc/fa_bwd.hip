@@ -125,7 +125,6 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dk, bf16* __restrict__ dv,
-    float* __restrict__ dkf, float* __restrict__ dvf,
     int S, int H, int HKV, int causal,
     long qsb, long qsh, long qss, long ksb, long ksh, long kss,
     long vsb, long vsh, long vss, long dosb, long dosh, long doss) {
@@ -142,13 +141,6 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   const int hkv = bh % HKV;
   const int b = bh / HKV;
   const int G = H / HKV;
-  // GQA q-heads are SPLIT ACROSS BLOCKS (blockIdx.z): with the head loop
-  // inside one block, the causal work imbalance (kv=0 blocks sweep every
-  // q tile, high-kv blocks almost none) on an exactly-chip-filling grid
-  // left the tail running on <10% of the CUs (PMC: avg concurrency 157
-  // waves vs dq's 309 — gpurun_out/prof3).  Finer blocks re-pack; dk/dv
-  // accumulate via fp32 atomics when G > 1 (one add per element per head).
-  const int gh = blockIdx.z;
   const int kvbase = blockIdx.x * KVROWS;
   const int kvw0 = kvbase + wave * 32;             // wave's first kv row
   const int kvrow = min(kvw0 + lq, S - 1);         // lane's kv row (clamped)
@@ -254,17 +246,22 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   };
 
   const int q0 = causal ? min(kvbase & ~(FB_QB - 1), S) : 0;
-  const int h = hkv * G + gh;
-  const bf16* qp = q + (long)b * qsb + (long)h * qsh;
-  const bf16* dop = dout + (long)b * dosb + (long)h * dosh;
-  const float* lp = lse + ((long)b * H + h) * S;
-  const float* dp = delta + ((long)b * H + h) * S;
-  load_rm(qp, dop, qss, doss, q0);
+  {
+    const int h0 = hkv * G;
+    load_rm(q + (long)b * qsb + (long)h0 * qsh,
+            dout + (long)b * dosb + (long)h0 * dosh, qss, doss, q0);
+  }
   write_rm();
   __syncthreads();
   build_tr();
 
-  {
+  for (int gh = 0; gh < G; ++gh) {
+    const int h = hkv * G + gh;
+    const bf16* qp = q + (long)b * qsb + (long)h * qsh;
+    const bf16* dop = dout + (long)b * dosb + (long)h * dosh;
+    const float* lp = lse + ((long)b * H + h) * S;
+    const float* dp = delta + ((long)b * H + h) * S;
+
     for (int qb = q0; qb < S; qb += FB_QB) {
       if (tid < FB_QB) {
         const int grow = min(qb + tid, S - 1);
@@ -273,8 +270,10 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
       }
       __syncthreads();
 
-      const bool has_next = qb + FB_QB < S;
-      const int qb_n = qb + FB_QB;
+      // next tile in the (gh, qb) sequence, for the prefetch
+      const int qb_n = qb + FB_QB < S ? qb + FB_QB : q0;
+      const int gh_n = qb + FB_QB < S ? gh : gh + 1;
+      const bool has_next = gh_n < G;
 
       // ---- per 32-row q sub-tile: C[q, kv] products, elementwise, repack,
       // dV^T/dK^T accumulate
@@ -303,7 +302,9 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
         __builtin_amdgcn_s_setprio(0);
         if (ct == 0 && has_next) {
           // issue next tile's global loads under the remaining compute
-          load_rm(qp, dop, qss, doss, qb_n);
+          const int hh = hkv * G + gh_n;
+          load_rm(q + (long)b * qsb + (long)hh * qsh,
+                  dout + (long)b * dosb + (long)hh * dosh, qss, doss, qb_n);
         }
         // overwrite sacc/dpacc in place with P / dS (register economy)
         const int kvg = kvw0 + lq;                 // lane's kv column
@@ -348,42 +349,21 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
     }
   }
 
-  // ---- epilogue: C[d, kv] (lane = kv column) -> dk/dv[kv][d]
-  // (dk/dv dense [B,HKV,S,D]; bf16 direct stores at G==1, fp32 atomics
-  // across the G q-head blocks otherwise)
+  // ---- epilogue: C[d, kv] (lane = kv column) -> dk/dv[kv][d] bf16
   const int kvg = kvw0 + lq;
   if (kvg < S) {
-    const long base = (((long)b * HKV + hkv) * S + kvg) * DH;
-    if (G == 1) {
-      bf16* dkp = dk + base;
-      bf16* dvp = dv + base;
+    bf16* dkp = dk + (((long)b * HKV + hkv) * S + kvg) * DH;
+    bf16* dvp = dv + (((long)b * HKV + hkv) * S + kvg) * DH;
+    // (dk/dv are allocated contiguous [B,HKV,S,D] by the host wrapper)
+    #pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
       #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        #pragma unroll
-        for (int r = 0; r < 16; r += 2) {
-          const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
-          *reinterpret_cast<unsigned int*>(dkp + d) =
-              fb_pk(dkacc[dt][r], dkacc[dt][r + 1]);
-          *reinterpret_cast<unsigned int*>(dvp + d) =
-              fb_pk(dvacc[dt][r], dvacc[dt][r + 1]);
-        }
-      }
-    } else {
-      // per-head fp32 partials at [gh][B,HKV,S,D]; summed host-side
-      const long pbase = (long)gh * (long)gridDim.y * S * DH +
-                         base;  // gridDim.y == B*HKV
-      float* dkp = dkf + pbase;
-      float* dvp = dvf + pbase;
-      #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        #pragma unroll
-        for (int r = 0; r < 16; r += 2) {
-          const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
-          *reinterpret_cast<float2*>(dkp + d) =
-              make_float2(dkacc[dt][r], dkacc[dt][r + 1]);
-          *reinterpret_cast<float2*>(dvp + d) =
-              make_float2(dvacc[dt][r], dvacc[dt][r + 1]);
-        }
+      for (int r = 0; r < 16; r += 2) {
+        const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
+        *reinterpret_cast<unsigned int*>(dkp + d) =
+            fb_pk(dkacc[dt][r], dkacc[dt][r + 1]);
+        *reinterpret_cast<unsigned int*>(dvp + d) =
+            fb_pk(dvacc[dt][r], dvacc[dt][r + 1]);
       }
     }
   }
@@ -622,23 +602,12 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   dim3 grid_kv128((S + 127) / 128, B * HKV);
   dim3 grid_kv((S + 255) / 256, B * HKV);
   dim3 grid_q((S + 255) / 256, B * H);
-  // GQA (G > 1): q-heads are separate blocks accumulating via fp32 atomics
-  const int G = H / HKV;
-  at::Tensor dkf, dvf;
-  if (G > 1) {
-    // per-head partials, plain stores (atomics measured 12% slower overall)
-    dkf = at::empty({G, B, HKV, S, DH}, q.options().dtype(at::kFloat));
-    dvf = at::empty({G, B, HKV, S, DH}, q.options().dtype(at::kFloat));
-  }
   const auto LKV = [&](auto kern, dim3 g, dim3 blk) {
-    g.z = G;
     hipLaunchKernelGGL(kern, g, blk, 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                        (const bf16*)v.data_ptr(), (const bf16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(),
-                       G > 1 ? dkf.data_ptr<float>() : nullptr,
-                       G > 1 ? dvf.data_ptr<float>() : nullptr,
                        S, H, HKV, causal ? 1 : 0,
                        q.stride(0), q.stride(1), q.stride(2),
                        k.stride(0), k.stride(1), k.stride(2),
@@ -662,10 +631,6 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   } else {
     LKV(fa_bwd_dkv_v1<64, 8>, grid_kv, dim3(512));
     LQ(fa_bwd_dq_v1<64>);
-  }
-  if (G > 1) {
-    dk.copy_(dkf.sum(0));
-    dv.copy_(dvf.sum(0));
   }
   return {dq, dk, dv};
 }
