@@ -40,6 +40,7 @@ class FusedAdamW(torch.optim.Optimizer):
         weight_decay: float = 1e-2,
         adam_w_mode: bool = True,
         offload_state: bool = False,
+        offload_path: Optional[str] = None,
     ):
         """``offload_state=True`` keeps exp_avg/exp_avg_sq (and the fp32
         master for bf16 params) in pinned host memory — the in-house
@@ -47,7 +48,13 @@ class FusedAdamW(torch.optim.Optimizer):
         (``DeepspeedOffloadOptimizerConfig``, reference ``configs.py:308-342``).
         Each step streams state H2D, runs the fused HIP kernel, and streams
         it back D2H (pinned + non_blocking, ordered on the current stream);
-        HBM then only holds params+grads, trading step time for capacity."""
+        HBM then only holds params+grads, trading step time for capacity.
+
+        ``offload_path`` additionally backs the state with FILES under that
+        directory (``torch.from_file`` shared mappings) — the NVMe offload
+        tier (``device="nvme"`` + ``nvme_path``, reference
+        ``configs.py:308-342``): state pages live in the page cache and
+        spill to disk under memory pressure instead of pinning host RAM."""
         if lr < 0.0:
             raise ValueError(f"Invalid learning rate: {lr}")
         defaults = dict(
@@ -55,7 +62,31 @@ class FusedAdamW(torch.optim.Optimizer):
             adam_w_mode=adam_w_mode,
         )
         self.offload_state = offload_state
+        self.offload_path = offload_path
+        if offload_path is not None:
+            import os
+
+            os.makedirs(offload_path, exist_ok=True)
+            self._offload_seq = 0
         super().__init__(params, defaults)
+
+    def _host_state_tensor(self, p) -> torch.Tensor:
+        """fp32 host tensor for one state slot: pinned RAM, or a shared
+        file mapping under offload_path (NVMe tier)."""
+        if self.offload_path is None:
+            return torch.zeros(
+                p.shape, dtype=torch.float32, device="cpu",
+                pin_memory=p.is_cuda,
+            )
+        import os
+
+        fn = os.path.join(self.offload_path,
+                          f"adamw_state_{self._offload_seq}.bin")
+        self._offload_seq += 1
+        t = torch.from_file(fn, shared=True, size=p.numel(),
+                            dtype=torch.float32).view(p.shape)
+        t.zero_()
+        return t
 
     @torch.no_grad()
     def step(
@@ -134,17 +165,16 @@ class FusedAdamW(torch.optim.Optimizer):
                 state = self.state[p]
                 if len(state) == 0:
                     state["step"] = 0
-                    state["exp_avg"] = torch.zeros(
-                        p.shape, dtype=torch.float32, device="cpu",
-                        pin_memory=on_gpu,
-                    )
-                    state["exp_avg_sq"] = torch.zeros(
-                        p.shape, dtype=torch.float32, device="cpu",
-                        pin_memory=on_gpu,
-                    )
+                    state["exp_avg"] = self._host_state_tensor(p)
+                    state["exp_avg_sq"] = self._host_state_tensor(p)
                     if p.dtype == torch.bfloat16:
-                        m = p.detach().float().cpu()
-                        state["master"] = m.pin_memory() if on_gpu else m
+                        if self.offload_path is not None:
+                            w = self._host_state_tensor(p)
+                            w.copy_(p.detach().float().cpu())
+                            state["master"] = w
+                        else:
+                            m = p.detach().float().cpu()
+                            state["master"] = m.pin_memory() if on_gpu else m
                 state["step"] += 1
                 m_dev = state["exp_avg"].to(p.device, non_blocking=True)
                 v_dev = state["exp_avg_sq"].to(p.device, non_blocking=True)

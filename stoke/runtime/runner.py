@@ -307,9 +307,20 @@ class StokeRunner:
                         f"{getattr(optimizer, '__name__', optimizer)}"
                     )
             elif dev == "nvme":
-                raise ValueError(
-                    "Stoke -- NVMe optimizer offload is not implemented"
-                )
+                import inspect
+
+                sig = inspect.signature(optimizer).parameters
+                if "offload_path" not in sig:
+                    raise ValueError(
+                        "Stoke -- NVMe optimizer-state offload requires an "
+                        "optimizer supporting offload_path (e.g. stoke "
+                        "FusedAdamW); got "
+                        f"{getattr(optimizer, '__name__', optimizer)}"
+                    )
+                nvme = off.nvme_path or "."
+                optimizer_kwargs = {**optimizer_kwargs,
+                                    "offload_state": True,
+                                    "offload_path": str(nvme)}
         if self._shard in ("oss", "sddp"):
             ocfg = self._status.oss_config
             opt = OSSOptimizer(

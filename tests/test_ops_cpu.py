@@ -199,3 +199,27 @@ def test_fused_sgd_scaler_skip_on_inf():
     assert torch.equal(p.detach(), torch.ones(8))
     opt.step(found_inf=torch.zeros(1))
     assert not torch.equal(p.detach(), torch.ones(8))
+
+
+def test_fused_adamw_nvme_offload(tmp_path):
+    """File-backed (NVMe-tier) state matches in-RAM offload/plain AdamW."""
+    import os
+
+    torch.manual_seed(0)
+    init = torch.randn(257)
+    p_a = torch.nn.Parameter(init.clone())
+    p_b = torch.nn.Parameter(init.clone())
+    o_a = FusedAdamW([p_a], lr=1e-2, weight_decay=0.01,
+                     offload_state=True, offload_path=str(tmp_path))
+    o_b = torch.optim.AdamW([p_b], lr=1e-2, weight_decay=0.01)
+    for step in range(4):
+        torch.manual_seed(step)
+        g = torch.randn_like(init)
+        p_a.grad = g.clone()
+        p_b.grad = g.clone()
+        o_a.step()
+        o_b.step()
+    assert torch.allclose(p_a, p_b, rtol=1e-5, atol=1e-7)
+    # state really is file-backed
+    files = [f for f in os.listdir(tmp_path) if f.startswith("adamw_state_")]
+    assert len(files) == 2  # exp_avg + exp_avg_sq for the fp32 param
