@@ -141,17 +141,25 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   const int hkv = bh % HKV;
   const int b = bh / HKV;
   const int G = H / HKV;
-  const int kvbase = blockIdx.x * KVROWS;
-  const int kvw0 = kvbase + wave * 32;             // wave's first kv row
-  const int kvrow = min(kvw0 + lq, S - 1);         // lane's kv row (clamped)
-
+  // Causal load-balancing: the host launches HALF the kv strips and each
+  // block processes strip b AND its mirror (nstrip-1-b) back to back, so
+  // every block sweeps ~the same number of q tiles (a kv=0 block alone
+  // sweeps 32x more than a kv=max one — PMC showed the tail running at
+  // <10% concurrency on the naive one-strip-per-block grid).
+  const int nstrip = (S + KVROWS - 1) / KVROWS;
   const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
   const bf16* vp = v + (long)b * vsb + (long)hkv * vsh;
 
+  for (int halfi = 0; halfi < 2; ++halfi) {
+    const int strip = halfi == 0 ? (int)blockIdx.x
+                                 : nstrip - 1 - (int)blockIdx.x;
+    if (halfi == 1 && (!causal || strip <= (int)blockIdx.x)) break;
+    const int kvbase = strip * KVROWS;
+    const int kvw0 = kvbase + wave * 32;           // wave's first kv row
+    const int kvrow = min(kvw0 + lq, S - 1);       // lane's kv row (clamped)
+
   // Wave-private K/V fragments (the 32x32x16 A and B lane maps are
   // identical, so these registers serve as the B operand directly).
-  // D=128 runs 4 waves x waves_per_eu(1) so each wave owns the whole
-  // 512-VGPR file — resident K/V + 128 accumulator registers, no spill.
   bf16x8v kf[DCH], vf[DCH];
   #pragma unroll
   for (int ch = 0; ch < DCH; ++ch) {
@@ -367,6 +375,8 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
       }
     }
   }
+  __syncthreads();  // LDS images reused by the mirror strip
+  }  // halfi
 }
 
 // ---------------------------------------------------------------------------
@@ -391,14 +401,20 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
   const int h = bh % H;
   const int hkv = h / (H / HKV);
   const int b = bh / H;
-  const int qbase = blockIdx.x * 256;
-  const int row0 = qbase + wave * 32;
-  const int qrow = min(row0 + lq, S - 1);
-
+  // Causal load balance: block handles q strip b AND its mirror (cf. dkv)
+  const int nstrip = (S + 255) / 256;
   const bf16* qp = q + (long)b * qsb + (long)h * qsh;
   const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
   const bf16* vp = v + (long)b * vsb + (long)hkv * vsh;
   const bf16* dop = dout + (long)b * dosb + (long)h * dosh;
+
+  for (int halfi = 0; halfi < 2; ++halfi) {
+  const int strip = halfi == 0 ? (int)blockIdx.x
+                               : nstrip - 1 - (int)blockIdx.x;
+  if (halfi == 1 && (!causal || strip <= (int)blockIdx.x)) break;
+  const int qbase = strip * 256;
+  const int row0 = qbase + wave * 32;
+  const int qrow = min(row0 + lq, S - 1);
   const float Lq = lse[((long)b * H + h) * S + qrow];
   const float Dq = delta[((long)b * H + h) * S + qrow];
 
@@ -564,6 +580,8 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
       }
     }
   }
+  __syncthreads();  // LDS images reused by the mirror strip
+  }  // halfi
 }
 
 }  // namespace
@@ -598,10 +616,13 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto dq = at::empty({B, H, S, DH}, q.options());
   auto dk = at::empty({B, HKV, S, DH}, q.options());
   auto dv = at::empty({B, HKV, S, DH}, q.options());
-  // D=128 dkv runs 4 waves/block (whole VGPR file per wave: spill-free)
-  dim3 grid_kv128((S + 127) / 128, B * HKV);
-  dim3 grid_kv((S + 255) / 256, B * HKV);
-  dim3 grid_q((S + 255) / 256, B * H);
+  // D=128 dkv runs 4 waves/block (whole VGPR file per wave: spill-free).
+  // Causal runs launch half the kv strips; each block also does its mirror.
+  const int ns128 = (S + 127) / 128, ns256 = (S + 255) / 256;
+  dim3 grid_kv128(causal ? (ns128 + 1) / 2 : ns128, B * HKV);
+  dim3 grid_kv(causal ? (ns256 + 1) / 2 : ns256, B * HKV);
+  const int nsq = (S + 255) / 256;
+  dim3 grid_q(causal ? (nsq + 1) / 2 : nsq, B * H);
   const auto LKV = [&](auto kern, dim3 g, dim3 blk) {
     hipLaunchKernelGGL(kern, g, blk, 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),

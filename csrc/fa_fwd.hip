@@ -128,9 +128,6 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   const int h = bh % H;
   const int hkv = h / (H / HKV);
   const int b = bh / H;
-  const int qbase = blockIdx.x * 256;
-  const int row0 = qbase + wave * 32;          // wave's first q row
-  const int qrow = min(row0 + lq, S - 1);      // this lane's q row (clamped)
 
   const bf16* qp = q + (long)b * qsb + (long)h * qsh;
   const bf16* kp = k + (long)b * ksb + (long)hkv * ksh;
@@ -140,6 +137,17 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
   // V image: transposed [DH][KVB + pad].
   __shared__ bf16 kbuf[FA_KVB * DH];
   __shared__ bf16 vbuf[DH * (FA_KVB + FA_VPAD)];
+
+  // Causal load balance: each block handles q strip b and its mirror
+  // (cf. fa_bwd; the kv sweep of strip 0 is 16x shorter than strip max)
+  const int nstrip = (S + 255) / 256;
+  for (int halfi = 0; halfi < 2; ++halfi) {
+  const int strip = halfi == 0 ? (int)blockIdx.x
+                               : nstrip - 1 - (int)blockIdx.x;
+  if (halfi == 1 && (!causal || strip <= (int)blockIdx.x)) break;
+  const int qbase = strip * 256;
+  const int row0 = qbase + wave * 32;          // wave's first q row
+  const int qrow = min(row0 + lq, S - 1);      // this lane's q row (clamped)
 
   // Q as B-operand fragments, resident for the whole KV sweep:
   // frag ch: lane holds q col lq, k-elems d = ch*16 + h2*8 + e.
@@ -358,6 +366,8 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     if (h2 == 0)
       lse[((long)b * H + h) * S + qg] = m + __logf(fmaxf(lsum, 1e-30f));
   }
+  __syncthreads();  // LDS tiles reused by the mirror strip
+  }  // halfi
 }
 
 }  // namespace
@@ -399,7 +409,8 @@ std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto out = out_bshd.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
-  dim3 grid((S + 255) / 256, B * H);
+  const int nstrip = (S + 255) / 256;
+  dim3 grid(causal ? (nstrip + 1) / 2 : nstrip, B * H);
   const auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(512), 0, stream,
                        (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
