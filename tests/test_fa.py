@@ -124,3 +124,48 @@ def test_fa_bwd_vs_sdpa(B, H, HKV, S, D, causal):
         err = (got.float() - want).abs().max().item()
         scale = want.abs().max().item() + 1e-6
         assert err / scale < 0.06, f"{nm} err {err} (scale {scale})"
+
+
+def test_fa_strided_views_match_contiguous():
+    """The model path feeds transposed projection views (BSHD storage);
+    the kernels take them with zero copies and must match contiguous."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from stoke.nn.attention import flash_attention
+
+    torch.manual_seed(5)
+    B, H, S, D = 2, 4, 512, 64
+    qkv = torch.randn(B, S, 3 * H * D, device="cuda").bfloat16()
+    q, k, v = qkv.split(H * D, dim=-1)
+    qv = q.view(B, S, H, D).transpose(1, 2)   # strided [B,H,S,D] view
+    kv_ = k.view(B, S, H, D).transpose(1, 2)
+    vv = v.view(B, S, H, D).transpose(1, 2)
+    for t in (qv, kv_, vv):
+        assert not t.is_contiguous()
+    out_v = flash_attention(qv, kv_, vv, causal=True)
+    out_c = flash_attention(qv.contiguous(), kv_.contiguous(),
+                            vv.contiguous(), causal=True)
+    assert torch.equal(out_v.float(), out_c.float())
+    # backward through the views
+    qg = qv.detach().clone().requires_grad_(True)
+    kg = kv_.detach().clone().requires_grad_(True)
+    vg = vv.detach().clone().requires_grad_(True)
+    flash_attention(qg, kg, vg, causal=True).sum().backward()
+    assert all(t.grad is not None and torch.isfinite(t.grad).all()
+               for t in (qg, kg, vg))
+
+
+def test_fa_attention_wrapper_routes():
+    """attention() uses the native kernels for bf16 CUDA and SDPA elsewhere."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from stoke.nn.attention import attention
+
+    torch.manual_seed(6)
+    q = torch.randn(1, 2, 128, 64, device="cuda").bfloat16()
+    k = torch.randn(1, 2, 128, 64, device="cuda").bfloat16()
+    v = torch.randn(1, 2, 128, 64, device="cuda").bfloat16()
+    out = attention(q, k, v, causal=True)
+    want = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), k.float(), v.float(), is_causal=True)
+    assert (out.float() - want).abs().max().item() < 0.05

@@ -74,3 +74,36 @@ def test_fp8_delayed_linear_numerics():
     assert rel < 0.1, f"fp8 delayed fwd rel err {rel}"
     assert x.grad is not None and torch.isfinite(x.grad).all()
     assert lin.weight.grad is not None
+
+
+def test_fp8_training_stability():
+    """Loss decreases and scales stay finite over 30 delayed-scaling steps."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from stoke.nn.fp8 import fp8_available
+    from stoke.nn import convert_linears_to_fp8
+
+    if not fp8_available():
+        pytest.skip("fp8 unavailable")
+    torch.manual_seed(7)
+    import torch.nn as nn
+
+    m = nn.Sequential(nn.Linear(1024, 2048), nn.SiLU(),
+                      nn.Linear(2048, 1024)).cuda().bfloat16()
+    assert convert_linears_to_fp8(m) == 2
+    opt = torch.optim.AdamW(m.parameters(), lr=1e-3)
+    x = torch.randn(64, 1024, device="cuda").bfloat16()
+    y = torch.randn(64, 1024, device="cuda").bfloat16()
+    losses = []
+    for _ in range(30):
+        loss = (m(x) - y).float().pow(2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)).tolist())
+    assert losses[-1] < losses[0] * 0.7, losses[::6]
+    for mod in m.modules():
+        for name in ("_sx", "_sw", "_sdy"):
+            if hasattr(mod, name):
+                assert torch.isfinite(getattr(mod, name)).all()
