@@ -269,11 +269,13 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     // softmax + PV compute below (T14 async-stage split)
     if (kb + FA_KVB < kv_end) load_tile(kb + FA_KVB);
 
-    // ---- online softmax (lane-local rows; combine lane<->lane+32)
+    // ---- online softmax (lane-local rows; combine lane<->lane+32).
+    // defer-max (guide T13): when this tile's max is within 8 of the
+    // running max for EVERY row in the wave, keep the old max — P stays
+    // bounded by e^8 and the whole O-rescale pass is skipped.
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
-    const float mnew = fmaxf(m, tmax);
-    const float alpha = __expf(m - mnew);
-    m = mnew;
+    const bool nores = __all(tmax - m <= 8.0f) && m > -1e29f;
+    const float mnew = nores ? m : fmaxf(m, tmax);
     float psum = 0.f;
     #pragma unroll
     for (int i = 0; i < 32; ++i) {
@@ -282,11 +284,17 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
       psum += p;
     }
     psum += __shfl_xor(psum, 32);
-    lsum = lsum * alpha + psum;
-    #pragma unroll
-    for (int dt = 0; dt < DT; ++dt)
+    if (nores) {
+      lsum += psum;
+    } else {
+      const float alpha = __expf(m - mnew);
+      m = mnew;
+      lsum = lsum * alpha + psum;
       #pragma unroll
-      for (int r = 0; r < 16; ++r) oacc[dt][r] *= alpha;
+      for (int dt = 0; dt < DT; ++dt)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[dt][r] *= alpha;
+    }
 
     // ---- P -> bf16 B-fragments in registers.  Fragment kch: lane needs
     // kv = kch*16 + h2*8 + e; elements e<4 live in the h=0 lane of this q,
