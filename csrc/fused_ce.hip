@@ -1,0 +1,166 @@
+// Fused cross-entropy for large-vocab LM heads (MI355X gfx950).
+//
+// The eager path materializes fp32 logits (`logits.float()`, 2x the bf16
+// bytes), then runs separate softmax fwd/bwd + NLL kernels — measured 5.7%
+// of a GPT-2 step plus the cast traffic (profiles/gpt2 steady state).  This
+// kernel computes loss = logsumexp(x) - x[target] per row in ONE online
+// pass over bf16 logits (running max + rescaled sum, the flash-attention
+// trick), and the backward emits bf16 (softmax - onehot) * scale directly.
+//
+// Shapes: logits [N, V] bf16 row-contiguous, target [N] int64; mean
+// reduction over rows with target != ignore_index (torch semantics).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+typedef __hip_bfloat16 bf16;
+
+union ce_bf16x8 {
+  uint4 u4;
+  unsigned short h[8];
+};
+
+__device__ __forceinline__ float ce_b2f(unsigned short v) {
+  __hip_bfloat16_raw r;
+  r.x = v;
+  return __bfloat162float(*reinterpret_cast<bf16*>(&r));
+}
+
+// One block (256 threads) per row: online max/sum in a single sweep.
+__global__ __launch_bounds__(256) void ce_fwd_kernel(
+    const bf16* __restrict__ logits, const long* __restrict__ target,
+    float* __restrict__ lse, float* __restrict__ loss_sum,
+    int* __restrict__ n_valid, long N, long V, long ignore_index) {
+  const long row = blockIdx.x;
+  if (row >= N) return;
+  const bf16* x = logits + row * V;
+  const long t = target[row];
+  float m = -1e30f, s = 0.f;
+  const long v8 = V & ~7L;
+  for (long i = (long)threadIdx.x * 8; i < v8; i += 256L * 8) {
+    ce_bf16x8 vv;
+    vv.u4 = *reinterpret_cast<const uint4*>(x + i);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float f = ce_b2f(vv.h[e]);
+      if (f > m) {
+        s *= __expf(m - f);
+        m = f;
+      }
+      s += __expf(f - m);
+    }
+  }
+  for (long i = v8 + threadIdx.x; i < V; i += 256) {
+    const float f = ce_b2f(*reinterpret_cast<const unsigned short*>(x + i));
+    if (f > m) {
+      s *= __expf(m - f);
+      m = f;
+    }
+    s += __expf(f - m);
+  }
+  // combine the 256 partial (m, s) pairs: wave shuffle then LDS
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float mo = __shfl_down(m, off);
+    const float so = __shfl_down(s, off);
+    const float mn = fmaxf(m, mo);
+    s = s * __expf(m - mn) + so * __expf(mo - mn);
+    m = mn;
+  }
+  __shared__ float sm[4], ss[4];
+  if ((threadIdx.x & 63) == 0) {
+    sm[threadIdx.x >> 6] = m;
+    ss[threadIdx.x >> 6] = s;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    m = sm[0];
+    s = ss[0];
+    #pragma unroll
+    for (int w = 1; w < 4; ++w) {
+      const float mn = fmaxf(m, sm[w]);
+      s = s * __expf(m - mn) + ss[w] * __expf(sm[w] - mn);
+      m = mn;
+    }
+    const float L = m + __logf(fmaxf(s, 1e-30f));
+    lse[row] = L;
+    if (t != ignore_index) {
+      const float xt = ce_b2f(
+          *reinterpret_cast<const unsigned short*>(x + t));
+      atomicAdd(loss_sum, L - xt);
+      atomicAdd(n_valid, 1);
+    }
+  }
+}
+
+// dlogits[i,j] = (exp(x - lse_i) - (j == t_i)) * g  (0 for ignored rows)
+__global__ __launch_bounds__(256) void ce_bwd_kernel(
+    const bf16* __restrict__ logits, const long* __restrict__ target,
+    const float* __restrict__ lse, const float* __restrict__ gscale,
+    bf16* __restrict__ dlogits, long N, long V, long ignore_index) {
+  const long row = blockIdx.x;
+  if (row >= N) return;
+  const bf16* x = logits + row * V;
+  bf16* dx = dlogits + row * V;
+  const long t = target[row];
+  const float g = (t == ignore_index) ? 0.f : *gscale;
+  const float L = lse[row];
+  const long v8 = V & ~7L;
+  for (long i = (long)threadIdx.x * 8; i < v8; i += 256L * 8) {
+    ce_bf16x8 vv, ov;
+    vv.u4 = *reinterpret_cast<const uint4*>(x + i);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float p = __expf(ce_b2f(vv.h[e]) - L);
+      if (i + e == t) p -= 1.f;
+      bf16 o = __float2bfloat16(p * g);
+      ov.h[e] = *reinterpret_cast<unsigned short*>(&o);
+    }
+    *reinterpret_cast<uint4*>(dx + i) = ov.u4;
+  }
+  for (long i = v8 + threadIdx.x; i < V; i += 256) {
+    float p = __expf(ce_b2f(*reinterpret_cast<const unsigned short*>(x + i)) - L);
+    if (i == t) p -= 1.f;
+    dx[i] = __float2bfloat16(p * g);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target,
+                               int64_t ignore_index) {
+  TORCH_CHECK(logits.scalar_type() == at::kBFloat16 && logits.dim() == 2 &&
+                  logits.is_contiguous(),
+              "ce_fwd: contiguous 2-D bf16 logits");
+  TORCH_CHECK(target.scalar_type() == at::kLong && target.is_contiguous(),
+              "ce_fwd: contiguous int64 targets");
+  const long N = logits.size(0), V = logits.size(1);
+  auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+  auto loss_sum = at::zeros({1}, logits.options().dtype(at::kFloat));
+  auto n_valid = at::zeros({1}, logits.options().dtype(at::kInt));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(ce_fwd_kernel, dim3((unsigned)N), dim3(256), 0, stream,
+                     (const bf16*)logits.data_ptr(),
+                     target.data_ptr<long>(), lse.data_ptr<float>(),
+                     loss_sum.data_ptr<float>(), n_valid.data_ptr<int>(),
+                     N, V, ignore_index);
+  return {loss_sum, n_valid, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                  at::Tensor gscale, int64_t ignore_index) {
+  const long N = logits.size(0), V = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(ce_bwd_kernel, dim3((unsigned)N), dim3(256), 0, stream,
+                     (const bf16*)logits.data_ptr(),
+                     target.data_ptr<long>(), lse.data_ptr<float>(),
+                     gscale.data_ptr<float>(), (bf16*)dlogits.data_ptr(),
+                     N, V, ignore_index);
+  return dlogits;
+}

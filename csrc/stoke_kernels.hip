@@ -635,6 +635,12 @@ void amp_update_scale_(at::Tensor scale, at::Tensor growth_tracker,
                      (float)backoff_factor, (int)growth_interval);
 }
 
+// Fused cross-entropy (csrc/fused_ce.hip)
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target,
+                               int64_t ignore_index);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                  at::Tensor gscale, int64_t ignore_index);
+
 // Fused NHWC BatchNorm kernels (csrc/fused_bn.hip)
 std::vector<at::Tensor> bn_fwd_train(
     at::Tensor x, c10::optional<at::Tensor> residual, at::Tensor gamma,
@@ -685,6 +691,8 @@ std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
                                     at::Tensor amax_next, int64_t kind);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ce_fwd", &ce_fwd, "fused bf16 cross-entropy forward (online lse)");
+  m.def("ce_bwd", &ce_bwd, "fused bf16 cross-entropy backward");
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN forward (eval)");

@@ -74,10 +74,9 @@ def pad_collate(batch):
 
 
 def lm_loss(logits, target):
-    return nn.functional.cross_entropy(
-        logits.reshape(-1, logits.shape[-1]).float(), target.reshape(-1),
-        ignore_index=-100,
-    )
+    from stoke.nn import fused_cross_entropy
+
+    return fused_cross_entropy(logits, target, ignore_index=-100)
 
 
 def main():

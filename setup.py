@@ -38,7 +38,8 @@ setup(
                      "csrc/fused_layernorm.hip",
                      "csrc/fa_fwd.hip",
                      "csrc/fa_bwd.hip",
-                     "csrc/fp8_quant.hip"],
+                     "csrc/fp8_quant.hip",
+                     "csrc/fused_ce.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

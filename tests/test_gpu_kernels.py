@@ -310,3 +310,31 @@ def test_multi_tensor_many_tensors():
     for t, r in zip(ts, refs):
         assert torch.allclose(t, r * 0.5)
     assert found.item() == 0.0
+
+
+def test_fused_cross_entropy_vs_torch():
+    """Fused bf16 online-lse CE vs fp32 torch, values and gradients."""
+    from stoke.nn import fused_cross_entropy
+
+    torch.manual_seed(11)
+    for N, V in [(64, 50257), (33, 1031), (8, 8)]:
+        logits = (torch.randn(N, V, device="cuda") * 3).bfloat16()
+        target = torch.randint(0, V, (N,), device="cuda")
+        target[::5] = -100  # ignored positions
+        a = logits.clone().requires_grad_(True)
+        b = logits.float().detach().requires_grad_(True)
+        la = fused_cross_entropy(a, target)
+        lb = torch.nn.functional.cross_entropy(b, target, ignore_index=-100)
+        assert abs(la.item() - lb.item()) / (abs(lb.item()) + 1e-6) < 2e-2, \
+            (N, V, la.item(), lb.item())
+        (la * 3.0).backward()
+        (lb * 3.0).backward()
+        err = (a.grad.float() - b.grad).abs().max().item()
+        scale = b.grad.abs().max().item() + 1e-9
+        assert err / scale < 0.08, (N, V, err, scale)
+    # all-ignored edge: loss 0, grads 0
+    t2 = torch.full((16,), -100, device="cuda", dtype=torch.long)
+    l2 = torch.randn(16, 128, device="cuda").bfloat16().requires_grad_(True)
+    out = fused_cross_entropy(l2, t2)
+    out.backward()
+    assert out.item() == 0.0 and torch.all(l2.grad == 0)
