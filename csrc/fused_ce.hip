@@ -41,7 +41,8 @@ __global__ __launch_bounds__(256) void ce_fwd_kernel(
   const bf16* x = logits + row * V;
   const long t = target[row];
   float m = -1e30f, s = 0.f;
-  const long v8 = V & ~7L;
+  // vectorized path only when every row base stays 16-B aligned
+  const long v8 = (V % 8 == 0) ? (V & ~7L) : 0;
   for (long i = (long)threadIdx.x * 8; i < v8; i += 256L * 8) {
     ce_bf16x8 vv;
     vv.u4 = *reinterpret_cast<const uint4*>(x + i);
@@ -110,7 +111,7 @@ __global__ __launch_bounds__(256) void ce_bwd_kernel(
   const long t = target[row];
   const float g = (t == ignore_index) ? 0.f : *gscale;
   const float L = lse[row];
-  const long v8 = V & ~7L;
+  const long v8 = (V % 8 == 0) ? (V & ~7L) : 0;
   for (long i = (long)threadIdx.x * 8; i < v8; i += 256L * 8) {
     ce_bf16x8 vv, ov;
     vv.u4 = *reinterpret_cast<const uint4*>(x + i);
@@ -144,6 +145,7 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target,
   auto loss_sum = at::zeros({1}, logits.options().dtype(at::kFloat));
   auto n_valid = at::zeros({1}, logits.options().dtype(at::kInt));
   auto stream = at::hip::getCurrentHIPStream().stream();
+  if (N == 0) return {loss_sum, n_valid, lse};
   hipLaunchKernelGGL(ce_fwd_kernel, dim3((unsigned)N), dim3(256), 0, stream,
                      (const bf16*)logits.data_ptr(),
                      target.data_ptr<long>(), lse.data_ptr<float>(),
@@ -157,6 +159,7 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
   const long N = logits.size(0), V = logits.size(1);
   auto dlogits = at::empty_like(logits);
   auto stream = at::hip::getCurrentHIPStream().stream();
+  if (N == 0) return dlogits;
   hipLaunchKernelGGL(ce_bwd_kernel, dim3((unsigned)N), dim3(256), 0, stream,
                      (const bf16*)logits.data_ptr(),
                      target.data_ptr<long>(), lse.data_ptr<float>(),
