@@ -167,3 +167,65 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
                      N, V, ignore_index);
   return dlogits;
 }
+
+// ---------------------------------------------------------------------------
+// Column sum for bias gradients: dY [N, C] bf16 row-major -> db [C].
+// torch's generic reduce ran this at 1.3 TB/s (16% of HBM — column-major
+// access pattern); this kernel streams row-major 128-B lines: each block
+// owns a 64-column stripe and a slice of rows, accumulates in fp32
+// registers, one atomicAdd per column per block.
+// ---------------------------------------------------------------------------
+namespace {
+
+__global__ __launch_bounds__(256) void colsum_kernel(
+    const bf16* __restrict__ dy, float* __restrict__ acc, long N, long C) {
+  const long c0 = (long)blockIdx.x * 64;          // column stripe
+  const int lc = threadIdx.x & 63;                // col within stripe
+  const int lr = threadIdx.x >> 6;                // row lane (4 rows/iter)
+  const long rows_per = (N + gridDim.y - 1) / gridDim.y;
+  const long r0 = blockIdx.y * rows_per;
+  const long r1 = min(r0 + rows_per, N);
+  const long c = c0 + lc;
+  const bool active = c < C;  // no early return: the barrier must be uniform
+  float s = 0.f;
+  if (active)
+    for (long r = r0 + lr; r < r1; r += 4)
+      s += ce_b2f(*reinterpret_cast<const unsigned short*>(&dy[r * C + c]));
+  // combine the 4 row-lanes of this column (lanes c, c+64, c+128, c+192)
+  __shared__ float sm[256];
+  sm[threadIdx.x] = s;
+  __syncthreads();
+  if (lr == 0 && active) {
+    s = sm[lc] + sm[lc + 64] + sm[lc + 128] + sm[lc + 192];
+    atomicAdd(&acc[c], s);
+  }
+}
+
+__global__ void colsum_cast_kernel(const float* __restrict__ acc,
+                                   bf16* __restrict__ out, long C) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i < C) out[i] = __float2bfloat16(acc[i]);
+}
+
+}  // namespace
+
+at::Tensor colsum_bf16(at::Tensor dy) {
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 && dy.dim() == 2 &&
+                  dy.is_contiguous(),
+              "colsum_bf16: contiguous 2-D bf16");
+  const long N = dy.size(0), C = dy.size(1);
+  auto acc = at::zeros({C}, dy.options().dtype(at::kFloat));
+  auto out = at::empty({C}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  if (N > 0) {
+    const int splits = (int)std::min<long>((N + 511) / 512, 64);
+    dim3 grid((unsigned)((C + 63) / 64), splits);
+    hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, stream,
+                       (const bf16*)dy.data_ptr(), acc.data_ptr<float>(),
+                       N, C);
+  }
+  hipLaunchKernelGGL(colsum_cast_kernel, dim3((unsigned)((C + 255) / 256)),
+                     dim3(256), 0, stream, acc.data_ptr<float>(),
+                     (bf16*)out.data_ptr(), C);
+  return out;
+}
