@@ -255,7 +255,11 @@ class StokePerLossScaler(StokeGradScaler):
             for p, s in zip(params, stash):
                 if s is not None:
                     p.grad = s if p.grad is None else p.grad.add_(s)
-        state["found_inf"] = found
+        # OR with earlier micro-batches (gradient accumulation): an inf
+        # already folded into p.grad must still skip the step
+        prev = state["found_inf"]
+        state["found_inf"] = found if prev is None else torch.maximum(
+            prev.to(found.device), found)
         state["unscaled"] = True  # grads are already true units
 
     def unscale_(self, optimizer):

@@ -186,3 +186,22 @@ def test_git_version():
 
     v = get_versions()
     assert v["version"] and stoke.__version__ == v["version"]
+
+
+def test_per_loss_scaler_accum_keeps_found_inf():
+    """An inf folded into grads by micro-batch 1 must still skip the step
+    even when micro-batch 2 is clean (gradient accumulation)."""
+    import torch
+    from stoke.amp import StokePerLossScaler
+
+    p = torch.nn.Parameter(torch.ones(4))
+    opt = torch.optim.SGD([p], lr=0.1)
+    sc = StokePerLossScaler(init_scale=4.0, device="cpu")
+    bad = (p * torch.ones(4)).sum() * float("inf")
+    sc.backward_per_loss([bad], opt, [p])
+    good = (p * torch.ones(4)).sum()
+    sc.backward_per_loss([good], opt, [p])
+    before = p.detach().clone()
+    sc.step(opt)
+    sc.update()
+    assert torch.equal(p.detach(), before), "step ran over inf-tainted grads"
