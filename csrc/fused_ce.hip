@@ -179,25 +179,41 @@ namespace {
 
 __global__ __launch_bounds__(256) void colsum_kernel(
     const bf16* __restrict__ dy, float* __restrict__ acc, long N, long C) {
-  const long c0 = (long)blockIdx.x * 64;          // column stripe
-  const int lc = threadIdx.x & 63;                // col within stripe
-  const int lr = threadIdx.x >> 6;                // row lane (4 rows/iter)
-  const long rows_per = (N + gridDim.y - 1) / gridDim.y;
-  const long r0 = blockIdx.y * rows_per;
+  // Each block streams FULL rows of its row slice (perfectly coalesced
+  // 16-B loads covering 2048 columns per pass); every thread owns fixed
+  // column octets accumulated in registers, 8 atomics per octet at the end.
+  constexpr int MAXCH = 8;                        // supports C <= 16384
+  const int nchunk = (int)((C + 2047) / 2048);
+  const long rows_per = (N + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per;
   const long r1 = min(r0 + rows_per, N);
-  const long c = c0 + lc;
-  const bool active = c < C;  // no early return: the barrier must be uniform
-  float s = 0.f;
-  if (active)
-    for (long r = r0 + lr; r < r1; r += 4)
-      s += ce_b2f(*reinterpret_cast<const unsigned short*>(&dy[r * C + c]));
-  // combine the 4 row-lanes of this column (lanes c, c+64, c+128, c+192)
-  __shared__ float sm[256];
-  sm[threadIdx.x] = s;
-  __syncthreads();
-  if (lr == 0 && active) {
-    s = sm[lc] + sm[lc + 64] + sm[lc + 128] + sm[lc + 192];
-    atomicAdd(&acc[c], s);
+  float s[MAXCH][8];
+  #pragma unroll
+  for (int ch = 0; ch < MAXCH; ++ch)
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) s[ch][k] = 0.f;
+  const bool vec = (C % 8 == 0);
+  for (long r = r0; r < r1; ++r) {
+    const bf16* row = dy + r * C;
+    for (int ch = 0; ch < nchunk; ++ch) {
+      const long col = (long)ch * 2048 + (long)threadIdx.x * 8;
+      if (col + 8 <= C && vec) {
+        ce_bf16x8 v;
+        v.u4 = *reinterpret_cast<const uint4*>(row + col);
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) s[ch][k] += ce_b2f(v.h[k]);
+      } else {
+        for (int k = 0; k < 8 && col + k < C; ++k)
+          s[ch][k] += ce_b2f(
+              *reinterpret_cast<const unsigned short*>(row + col + k));
+      }
+    }
+  }
+  for (int ch = 0; ch < nchunk; ++ch) {
+    const long col = (long)ch * 2048 + (long)threadIdx.x * 8;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k)
+      if (col + k < C) atomicAdd(&acc[col + k], s[ch][k]);
   }
 }
 
@@ -217,10 +233,11 @@ at::Tensor colsum_bf16(at::Tensor dy) {
   auto acc = at::zeros({C}, dy.options().dtype(at::kFloat));
   auto out = at::empty({C}, dy.options());
   auto stream = at::hip::getCurrentHIPStream().stream();
+  TORCH_CHECK(C <= 16384, "colsum_bf16: C <= 16384");
   if (N > 0) {
-    const int splits = (int)std::min<long>((N + 511) / 512, 64);
-    dim3 grid((unsigned)((C + 63) / 64), splits);
-    hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, stream,
+    // enough row-slice blocks to fill the chip, each streaming whole rows
+    const int blocks = (int)std::min<long>((N + 15) / 16, 1024);
+    hipLaunchKernelGGL(colsum_kernel, dim3(blocks), dim3(256), 0, stream,
                        (const bf16*)dy.data_ptr(), acc.data_ptr<float>(),
                        N, C);
   }
