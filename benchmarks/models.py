@@ -143,15 +143,15 @@ class GPT2Block(nn.Module):
 
     def __init__(self, d, nh, dropout=0.0):
         super().__init__()
-        from stoke.nn import StokeLayerNorm, StokeLinear
+        from stoke.nn import StokeLayerNorm
 
         self.nh = nh
         self.ln1 = StokeLayerNorm(d)
-        self.qkv = StokeLinear(d, 3 * d)
-        self.proj = StokeLinear(d, d)
+        self.qkv = nn.Linear(d, 3 * d)
+        self.proj = nn.Linear(d, d)
         self.ln2 = StokeLayerNorm(d)
         self.mlp = nn.Sequential(
-            StokeLinear(d, 4 * d), nn.GELU(), StokeLinear(4 * d, d)
+            nn.Linear(d, 4 * d), nn.GELU(), nn.Linear(4 * d, d)
         )
 
     def forward(self, x):

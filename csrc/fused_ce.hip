@@ -168,88 +168,3 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
   return dlogits;
 }
 
-// ---------------------------------------------------------------------------
-// Column sum for bias gradients: dY [N, C] bf16 row-major -> db [C].
-// torch's generic reduce ran this at 1.3 TB/s (16% of HBM — column-major
-// access pattern); this kernel streams row-major 128-B lines: each block
-// owns a 64-column stripe and a slice of rows, accumulates in fp32
-// registers, one atomicAdd per column per block.
-// ---------------------------------------------------------------------------
-namespace {
-
-__global__ __launch_bounds__(256) void colsum_kernel(
-    const bf16* __restrict__ dy, float* __restrict__ acc, long N, long C) {
-  // Each block streams FULL rows of its row slice (perfectly coalesced
-  // 16-B loads covering 2048 columns per pass); every thread owns fixed
-  // column octets accumulated in registers, 8 atomics per octet at the end.
-  constexpr int MAXCH = 8;                        // supports C <= 16384
-  const int nchunk = (int)((C + 2047) / 2048);
-  const long rows_per = (N + gridDim.x - 1) / gridDim.x;
-  const long r0 = (long)blockIdx.x * rows_per;
-  const long r1 = min(r0 + rows_per, N);
-  float s[MAXCH][8];
-  #pragma unroll
-  for (int ch = 0; ch < MAXCH; ++ch)
-    #pragma unroll
-    for (int k = 0; k < 8; ++k) s[ch][k] = 0.f;
-  const bool vec = (C % 8 == 0);
-  for (long r = r0; r < r1; ++r) {
-    const bf16* row = dy + r * C;
-    for (int ch = 0; ch < nchunk; ++ch) {
-      const long col = (long)ch * 2048 + (long)threadIdx.x * 8;
-      if (col + 8 <= C && vec) {
-        ce_bf16x8 v;
-        v.u4 = *reinterpret_cast<const uint4*>(row + col);
-        #pragma unroll
-        for (int k = 0; k < 8; ++k) s[ch][k] += ce_b2f(v.h[k]);
-      } else {
-        for (int k = 0; k < 8 && col + k < C; ++k)
-          s[ch][k] += ce_b2f(
-              *reinterpret_cast<const unsigned short*>(row + col + k));
-      }
-    }
-  }
-  // plain per-block partial stores: atomics here measured 1024-way
-  // contention per address (GPT-2 dropped 237k -> 177k tok/s)
-  float* part = acc + (long)blockIdx.x * C;
-  for (int ch = 0; ch < nchunk; ++ch) {
-    const long col = (long)ch * 2048 + (long)threadIdx.x * 8;
-    #pragma unroll
-    for (int k = 0; k < 8; ++k)
-      if (col + k < C) part[col + k] = s[ch][k];
-  }
-}
-
-__global__ void colsum_cast_kernel(const float* __restrict__ part,
-                                   bf16* __restrict__ out, long C,
-                                   int nblocks) {
-  const long i = (long)blockIdx.x * 256 + threadIdx.x;
-  if (i >= C) return;
-  float s = 0.f;
-  for (int b = 0; b < nblocks; ++b) s += part[(long)b * C + i];
-  out[i] = __float2bfloat16(s);
-}
-
-}  // namespace
-
-at::Tensor colsum_bf16(at::Tensor dy) {
-  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 && dy.dim() == 2 &&
-                  dy.is_contiguous(),
-              "colsum_bf16: contiguous 2-D bf16");
-  const long N = dy.size(0), C = dy.size(1);
-  TORCH_CHECK(C <= 16384, "colsum_bf16: C <= 16384");
-  const int blocks =
-      N > 0 ? (int)std::min<long>((N + 15) / 16, 512) : 1;
-  auto part = at::zeros({(long)blocks, C}, dy.options().dtype(at::kFloat));
-  auto out = at::empty({C}, dy.options());
-  auto stream = at::hip::getCurrentHIPStream().stream();
-  if (N > 0) {
-    hipLaunchKernelGGL(colsum_kernel, dim3(blocks), dim3(256), 0, stream,
-                       (const bf16*)dy.data_ptr(), part.data_ptr<float>(),
-                       N, C);
-  }
-  hipLaunchKernelGGL(colsum_cast_kernel, dim3((unsigned)((C + 255) / 256)),
-                     dim3(256), 0, stream, part.data_ptr<float>(),
-                     (bf16*)out.data_ptr(), C, blocks);
-  return out;
-}

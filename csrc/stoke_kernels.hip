@@ -640,7 +640,6 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target,
                                int64_t ignore_index);
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
                   at::Tensor gscale, int64_t ignore_index);
-at::Tensor colsum_bf16(at::Tensor dy);
 
 // Fused NHWC BatchNorm kernels (csrc/fused_bn.hip)
 std::vector<at::Tensor> bn_fwd_train(
@@ -694,8 +693,6 @@ std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd, "fused bf16 cross-entropy forward (online lse)");
   m.def("ce_bwd", &ce_bwd, "fused bf16 cross-entropy backward");
-  m.def("colsum_bf16", &colsum_bf16,
-        "row-major streaming column sum (bias gradients)");
   m.def("bn_fwd_train", &bn_fwd_train,
         "fused NHWC bf16 BN forward (train): stats + scale/shift(+res)+relu");
   m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC bf16 BN forward (eval)");
