@@ -136,8 +136,9 @@ def main():
     elif args.model == "gpt2-oss":
         seq = args.seq or 1024
         model = models.gpt2_medium(max_seq=seq)
-        # 32/GPU measured 255.6k tok/s vs 224-237k at 16 (gpurun_out/call34)
-        batch = args.batch or 32
+        # 48/GPU measured 264.2k tok/s (255.6k at 32, 224-237k at 16 —
+        # gpurun_out/call34-35); llama b12 OOMs so its default stays 8
+        batch = args.batch or 48
         stoke_kw = dict(
             gpu=True,
             fp16="bf16",
