@@ -205,3 +205,38 @@ def test_per_loss_scaler_accum_keeps_found_inf():
     sc.step(opt)
     sc.update()
     assert torch.equal(p.detach(), before), "step ran over inf-tainted grads"
+
+
+def test_lazy_loss_arithmetic():
+    import torch
+    from stoke.utils import LazyLoss
+
+    a = LazyLoss(torch.tensor([6.0]))
+    b = LazyLoss(torch.tensor([2.0]))
+    assert float(a) == 6.0 and a.item() == 6.0
+    assert float(a + b) == 8.0 and float(a + 1.0) == 7.0 and float(1.0 + a) == 7.0
+    assert float(a - b) == 4.0 and float(10.0 - a) == 4.0
+    assert float(a * b) == 12.0 and float(0.5 * a) == 3.0
+    assert float(a / b) == 3.0 and float(12.0 / a) == 2.0
+    assert float(-a) == -6.0 and float(abs(LazyLoss(torch.tensor([-3.0])))) == 3.0
+    assert round(a / 4, 2) == 1.5
+    assert (a > b) and (b < a) and (a >= 6.0) and (a <= 6.0) and a == 6.0
+    assert f"{a:.3f}" == "6.000"
+    # deferred all-reduce semantics: div applied lazily at first use
+    c = LazyLoss(torch.tensor([9.0]), div=3.0)
+    assert float(c) == 3.0
+
+
+def test_fused_cross_entropy_cpu_fallback_matches_torch():
+    import torch
+    import torch.nn.functional as F
+    from stoke.nn import fused_cross_entropy
+
+    torch.manual_seed(0)
+    logits = torch.randn(3, 5, 17)
+    tgt = torch.randint(0, 17, (3, 5))
+    tgt[0, 0] = -100
+    a = fused_cross_entropy(logits, tgt, ignore_index=-100)
+    b = F.cross_entropy(logits.reshape(-1, 17).float(), tgt.reshape(-1),
+                        ignore_index=-100)
+    assert torch.allclose(a, b)
