@@ -338,3 +338,31 @@ def test_fused_cross_entropy_vs_torch():
     out = fused_cross_entropy(l2, t2)
     out.backward()
     assert out.item() == 0.0 and torch.all(l2.grad == 0)
+
+
+def test_per_loss_scaler_gpu():
+    """Per-loss scalers on CUDA: HIP stash-unscale path, independent backoff."""
+    from stoke.amp import StokePerLossScaler
+
+    p = torch.nn.Parameter(torch.ones(64, device="cuda"))
+    opt = torch.optim.SGD([p], lr=0.1)
+    sc = StokePerLossScaler(init_scale=2.0**8, device="cuda")
+    x = torch.ones(64, device="cuda")
+    a = (p * x).sum()
+    b = (p * x).sum() * float("inf")
+    sc.backward_per_loss([a, b], opt, [p])
+    before = p.detach().clone()
+    sc.step(opt)
+    sc.update()
+    torch.cuda.synchronize()
+    assert torch.equal(p.detach(), before)
+    assert sc._loss_scales[0].item() == 2.0**8
+    assert sc._loss_scales[1].item() == 2.0**7
+    opt.zero_grad()
+    sc.backward_per_loss([(p * x).sum(), (p * x).sum() * 2.0], opt, [p])
+    assert torch.allclose(p.grad, torch.full((64,), 3.0, device="cuda"),
+                          atol=1e-5)
+    sc.step(opt)
+    sc.update()
+    torch.cuda.synchronize()
+    assert not torch.equal(p.detach(), before)
