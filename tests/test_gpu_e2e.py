@@ -147,3 +147,35 @@ def test_lm_blocks_match_eager_reference():
     out2.float().pow(2).mean().backward()
     assert torch.isfinite(out2.float()).all()
     assert x2.grad is not None and torch.isfinite(x2.grad.float()).all()
+
+
+def test_tiny_gpt2_convergence():
+    """Full stack (FA + fused CE + fused LN + FusedAdamW through the Stoke
+    facade) overfits a fixed batch — catches any silent numerics break."""
+    from benchmarks import models
+    from stoke import Stoke, StokeOptimizer
+    from stoke.nn import fused_cross_entropy
+    from stoke.ops.fused_adam import FusedAdamW
+
+    torch.manual_seed(0)
+    model = models.GPT2(vocab=512, d=128, nlayer=2, nh=2, max_seq=128)
+    model = model.bfloat16()
+    s = Stoke(
+        model=model,
+        optimizer=StokeOptimizer(optimizer=FusedAdamW,
+                                 optimizer_kwargs={"lr": 3e-3}),
+        loss=lambda lg, t: fused_cross_entropy(lg, t),
+        batch_size_per_device=8,
+        gpu=True, fp16="bf16", verbose=False,
+    )
+    x = torch.randint(0, 512, (8, 128), device="cuda")
+    y = torch.randint(0, 512, (8, 128), device="cuda")
+    losses = []
+    for _ in range(60):
+        out = s.model(x)
+        loss = s.loss(out, y)
+        s.backward(loss)
+        s.step()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 0.5, losses[::12]
