@@ -185,7 +185,8 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
   __shared__ bf16 dotr[DH * (FB_QB + FB_PAD)];
   __shared__ float lsh[FB_QB], dsh[FB_QB];
 
-  const float scale = rsqrtf((float)DH);
+  const float scale = rsqrtf((float)DH);            // natural (dS math)
+  const float scale2 = scale * 1.44269504089f;      // exp2-domain exponent
   const int tid = threadIdx.x;
 
   // ---- tile pipeline (T14): the row-major Q/dO pieces for tile t+1 are
@@ -273,7 +274,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
     for (int qb = q0; qb < S; qb += FB_QB) {
       if (tid < FB_QB) {
         const int grow = min(qb + tid, S - 1);
-        lsh[tid] = lp[grow];
+        lsh[tid] = lp[grow] * 1.44269504089f;  // logsumexp in base-2 units
         dsh[tid] = dp[grow];
       }
       __syncthreads();
@@ -322,7 +323,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
           const int qg = qb + qloc;
           float p = 0.f;
           if (qg < S && kvg < S && (!causal || kvg <= qg))
-            p = __expf(sacc[r] * scale - lsh[qloc]);
+            p = exp2f(sacc[r] * scale2 - lsh[qloc]);
           const float ds = p * (dpacc[r] - dsh[qloc]) * scale;
           sacc[r] = p;
           dpacc[r] = ds;
@@ -415,7 +416,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
   const int qbase = strip * 256;
   const int row0 = qbase + wave * 32;
   const int qrow = min(row0 + lq, S - 1);
-  const float Lq = lse[((long)b * H + h) * S + qrow];
+  const float Lq = lse[((long)b * H + h) * S + qrow] * 1.44269504089f;
   const float Dq = delta[((long)b * H + h) * S + qrow];
 
   bf16x8v qf[DCH], dof[DCH];
@@ -436,7 +437,8 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
   __shared__ bf16 vrm[FB_QB * DH];
   __shared__ bf16 ktr[DH * (FB_QB + FB_PAD)];
 
-  const float scale = rsqrtf((float)DH);
+  const float scale = rsqrtf((float)DH);            // natural (dS math)
+  const float scale2 = scale * 1.44269504089f;      // exp2-domain exponent
   const int tid = threadIdx.x;
 
   // ---- tile pipeline (cf. fa_bwd_dkv_v1): K/V row-major pieces for tile
@@ -536,7 +538,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
         const int kvg = kb + ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
         float p = 0.f;
         if (kvg < S && qg < S && (!causal || kvg <= qg))
-          p = __expf(sacc[r] * scale - Lq);
+          p = exp2f(sacc[r] * scale2 - Lq);
         dpacc[r] = p * (dpacc[r] - Dq) * scale;
       }
       bf16x8v dsb[2];

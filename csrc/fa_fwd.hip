@@ -163,7 +163,9 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     #pragma unroll
     for (int r = 0; r < 16; ++r) oacc[dt][r] = 0.f;
   float m = -1e30f, lsum = 0.f;
-  const float scale = rsqrtf((float)DH);
+  // softmax runs in the exp2 domain (v_exp_f32 IS base-2; folding log2e
+  // into the scale removes one v_mul per exponential)
+  const float scale = rsqrtf((float)DH) * 1.44269504089f;
 
   // ---- tile staging (T14 register split: global loads for tile t+1 are
   // issued during tile t's compute; the LDS writes happen after the
@@ -279,7 +281,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     float psum = 0.f;
     #pragma unroll
     for (int i = 0; i < 32; ++i) {
-      const float p = __expf(p32[i] - mnew);
+      const float p = exp2f(p32[i] - mnew);
       p32[i] = p;
       psum += p;
     }
@@ -287,7 +289,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     if (nores) {
       lsum += psum;
     } else {
-      const float alpha = __expf(m - mnew);
+      const float alpha = exp2f(m - mnew);
       m = mnew;
       lsum = lsum * alpha + psum;
       #pragma unroll
@@ -371,8 +373,9 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
             pk_bf16(oacc[dt][r] * inv, oacc[dt][r + 1] * inv);
       }
     }
-    if (h2 == 0)
-      lse[((long)b * H + h) * S + qg] = m + __logf(fmaxf(lsum, 1e-30f));
+    if (h2 == 0)  // convert base-2 running max back to natural log for bwd
+      lse[((long)b * H + h) * S + qg] =
+          m * 0.69314718056f + __logf(fmaxf(lsum, 1e-30f));
   }
   __syncthreads();  // LDS tiles reused by the mirror strip
   }  // halfi
