@@ -323,7 +323,7 @@ __global__ __launch_bounds__(WAVES * 64) __attribute__((amdgpu_waves_per_eu(1, 2
           const int qg = qb + qloc;
           float p = 0.f;
           if (qg < S && kvg < S && (!causal || kvg <= qg))
-            p = exp2f(sacc[r] * scale2 - lsh[qloc]);
+            p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsh[qloc]);
           const float ds = p * (dpacc[r] - dsh[qloc]) * scale;
           sacc[r] = p;
           dpacc[r] = ds;
@@ -538,7 +538,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v1(
         const int kvg = kb + ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * h2;
         float p = 0.f;
         if (kvg < S && qg < S && (!causal || kvg <= qg))
-          p = exp2f(sacc[r] * scale2 - Lq);
+          p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - Lq);
         dpacc[r] = p * (dpacc[r] - Dq) * scale;
       }
       bf16x8v dsb[2];

@@ -281,7 +281,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     float psum = 0.f;
     #pragma unroll
     for (int i = 0; i < 32; ++i) {
-      const float p = exp2f(p32[i] - mnew);
+      const float p = __builtin_amdgcn_exp2f(p32[i] - mnew);
       p32[i] = p;
       psum += p;
     }
@@ -289,7 +289,7 @@ __global__ __launch_bounds__(512) void fa_fwd_v1(
     if (nores) {
       lsum += psum;
     } else {
-      const float alpha = exp2f(m - mnew);
+      const float alpha = __builtin_amdgcn_exp2f(m - mnew);
       m = mnew;
       lsum = lsum * alpha + psum;
       #pragma unroll
