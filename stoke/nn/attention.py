@@ -48,7 +48,7 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 def _fa_usable(q, k, v, causal) -> bool:
-    # Default ON: the strip-paired kernels measure 1.63x AOTriton fwd+bwd
+    # Default ON: the strip-paired kernels measure 1.64-1.66x AOTriton fwd+bwd
     # on both bench shapes (D=64 GPT-2 and D=128 GQA Llama —
     # benchmarks/fa_bench.py, NOTES.md).  STOKE_USE_FA=0 disables.
     if os.environ.get("STOKE_USE_FA", "1") == "0":
