@@ -414,6 +414,9 @@ std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const int HKV = k.size(1);
   TORCH_CHECK(DH == 64 || DH == 128, "fa_fwd: head dim 64 or 128");
   TORCH_CHECK(H % HKV == 0, "fa_fwd: H must be a multiple of H_kv");
+  TORCH_CHECK(k.size(2) == S && v.size(2) == S && k.size(3) == DH &&
+                  v.size(3) == DH && v.size(1) == HKV,
+              "fa_fwd: self-attention shapes only (shared S, D, H_kv)");
   // out in [B,S,H,D]-contiguous storage, returned as a [B,H,S,D] view:
   // the caller's transpose(1,2).reshape(B,S,H*D) is then a free view.
   auto out_bshd = at::empty({B, S, H, DH}, q.options());

@@ -58,6 +58,9 @@ def _fa_usable(q, k, v, causal) -> bool:
         return False
     if q.dim() != 4 or q.shape[-1] not in (64, 128):
         return False
+    # self-attention only: the kernels assume one shared sequence length
+    if k.shape[2] != q.shape[2] or v.shape[2] != q.shape[2]:
+        return False
     if k.shape[1] != q.shape[1] and q.shape[1] % k.shape[1] != 0:
         return False
     from stoke import ops
