@@ -90,7 +90,8 @@ __global__ __launch_bounds__(256) void ce_fwd_kernel(
     }
     const float L = m + __logf(fmaxf(s, 1e-30f));
     lse[row] = L;
-    if (t != ignore_index) {
+    // out-of-range targets are treated as ignored instead of reading OOB
+    if (t != ignore_index && t >= 0 && t < V) {
       const float xt = ce_b2f(
           *reinterpret_cast<const unsigned short*>(x + t));
       atomicAdd(loss_sum, L - xt);
@@ -109,7 +110,8 @@ __global__ __launch_bounds__(256) void ce_bwd_kernel(
   const bf16* x = logits + row * V;
   bf16* dx = dlogits + row * V;
   const long t = target[row];
-  const float g = (t == ignore_index) ? 0.f : *gscale;
+  const float g =
+      (t == ignore_index || t < 0 || t >= V) ? 0.f : *gscale;
   const float L = lse[row];
   const long v8 = (V % 8 == 0) ? (V & ~7L) : 0;
   for (long i = (long)threadIdx.x * 8; i < v8; i += 256L * 8) {
