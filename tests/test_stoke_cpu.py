@@ -263,3 +263,24 @@ def test_flops_profiler_counts():
     assert abs(fp.get_total_params() - sum(p.numel() for p in m.parameters())) < 1e4
     text = fp.print_model_profile(detailed=True)
     assert "GFLOPs" in text
+
+
+def test_flops_profiler_per_step_reset():
+    """ADVICE round-1 fix: with profile_step > 1 the printed profile covers
+    exactly ONE optimizer step's forwards, not everything since init."""
+    import torch.nn as nn
+    from stoke.utils import FlopsProfiler
+
+    model = nn.Linear(8, 4)
+    fp = FlopsProfiler(model)
+    fp.start_profile()
+    x = torch.randn(2, 8)
+    model(x)
+    model(x)
+    inflated = fp.get_total_flops()
+    fp.reset_flops()           # the facade calls this at step boundary
+    model(x)
+    per_step = fp.get_total_flops()
+    fp.stop_profile()
+    assert inflated == 2 * per_step
+    assert per_step == 2.0 * 2 * 4 * 8  # one forward: 2*N*out*in
