@@ -130,8 +130,8 @@ class DDPConfig:
 class DeepspeedAIOConfig:
     """Async-I/O knobs, accepted for compatibility (reference ``configs.py:191-219``).
 
-    NVMe offload is not implemented natively; these values are validated but
-    unused unless an offload device of "nvme" is requested (which raises).
+    The NVMe state tier uses OS page-cache-backed file mappings rather than
+    a userspace AIO engine, so these tuning values are accepted but unused.
     """
 
     block_size: int = 1048576
@@ -189,7 +189,9 @@ class DeepspeedOffloadOptimizerConfig:
     """Optimizer-state offload knobs (reference ``configs.py:308-342``).
 
     ``device='cpu'`` maps to pinned-host optimizer state with async HIP
-    H2D/D2H copies in the shard engine; ``'nvme'`` is unsupported and raises.
+    H2D/D2H copies; ``'nvme'`` maps to file-backed state tensors under
+    ``nvme_path`` (``torch.from_file`` shared mappings — page-cache hot,
+    spills to disk under memory pressure).
     """
 
     buffer_count: int = 4

@@ -219,3 +219,62 @@ def test_facade_deepspeed_zero1_gloo():
     torch.multiprocessing.spawn(
         _ds_worker, args=(2, free_port()), nprocs=2, join=True
     )
+
+
+def _ds_nvme_worker(rank, world, port, tmpdir):
+    os.environ.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), RANK=str(rank),
+        WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+    )
+    _patch_gpu_probes()
+    import torch.nn as nn
+
+    from stoke import (
+        DeepspeedConfig,
+        DeepspeedFP16Config,
+        DeepspeedOffloadOptimizerConfig,
+        DeepspeedZeROConfig,
+        Stoke,
+        StokeOptimizer,
+    )
+    from stoke.ops.fused_adam import FusedAdamW
+
+    torch.manual_seed(77)
+    s = Stoke(
+        model=nn.Linear(8, 4),
+        optimizer=StokeOptimizer(
+            optimizer=FusedAdamW, optimizer_kwargs={"lr": 1e-2}
+        ),
+        loss=nn.CrossEntropyLoss(),
+        batch_size_per_device=4,
+        gpu=True,
+        fp16="deepspeed",
+        distributed="deepspeed",
+        configs=[DeepspeedConfig(
+            dist_backend="gloo",
+            fp16=DeepspeedFP16Config(),
+            zero_optimization=DeepspeedZeROConfig(
+                stage=1,
+                offload_optimizer=DeepspeedOffloadOptimizerConfig(
+                    device="nvme", nvme_path=tmpdir),
+            ),
+        )],
+        verbose=False,
+    )
+    x = torch.randn(4, 8)
+    y = torch.randint(0, 4, (4,))
+    for _ in range(3):
+        out = s.model(x)
+        s.backward(s.loss(out, y))
+        s.step()
+    # state landed in files under nvme_path
+    files = [f for f in os.listdir(tmpdir) if f.startswith("adamw_state_")]
+    assert files, "NVMe-tier state files missing"
+    dist.destroy_process_group()
+
+
+def test_facade_deepspeed_nvme_offload_gloo(tmp_path):
+    torch.multiprocessing.spawn(
+        _ds_nvme_worker, args=(1, free_port(), str(tmp_path)), nprocs=1,
+        join=True,
+    )
