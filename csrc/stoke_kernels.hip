@@ -684,7 +684,7 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                at::Tensor out, at::Tensor dout,
                                at::Tensor lse, bool causal);
 
-// fp8 fused quantize — ROUND-2 WIP, env-gated tests only (csrc/fp8_quant.hip)
+// fp8 fused quantizers (csrc/fp8_quant.hip)
 at::Tensor fp8_quant(at::Tensor x, at::Tensor scale, at::Tensor amax_next,
                      int64_t kind);
 std::vector<at::Tensor> fp8_quant_t(at::Tensor x, at::Tensor scale,
@@ -710,15 +710,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe32", &mfma_probe32,
         "layout probe: one v_mfma_f32_32x32x16_bf16 (A[32,16] x B[16,32])");
   m.def("mfma_probe", &mfma_probe,
-        "16x16x32 bf16 MFMA fragment-layout probe (round-2 WIP)");
+        "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("fa_fwd", &fa_fwd,
-        "flash-attention forward v0 (round-2 WIP; not wired into models)");
+        "CDNA4 flash-attention forward (strip-paired FA-2, strided views)");
   m.def("fa_bwd", &fa_bwd,
-        "flash-attention backward v0 (round-2 WIP; not wired into models)");
+        "CDNA4 flash-attention backward (dkv + dq + delta)");
   m.def("fp8_quant", &fp8_quant,
-        "fused bf16->fp8 quantize + next-amax (round-2 WIP)");
+        "fused bf16->fp8 quantize + next-amax byproduct");
   m.def("fp8_quant_t", &fp8_quant_t,
-        "fused bf16->fp8 quantize emitting both layouts (round-2 WIP)");
+        "fused bf16->fp8 quantize emitting row-major AND transposed layouts");
   m.def("multi_tensor_unscale_", &multi_tensor_unscale_,
         "fused grad unscale + inf/nan check (HIP)");
   m.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq,
