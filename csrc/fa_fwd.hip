@@ -16,7 +16,7 @@
 //     v_cvt_pk_bf16_f32 pairs + one __shfl_xor(32) half-exchange per pair
 //     of dwords — no LDS staging of P.
 //
-// Fragment maps (validated on hardware by tests/test_fa_wip.py probes):
+// Fragment maps (validated on hardware by tests/test_fa.py probes):
 //   16x16x32: A row l&15, k=(l>>4)*8+e; B col l&15 same k;
 //             C col l&15, row (l>>4)*4+r.
 //   32x32x16: A row l&31, k=(l>>5)*8+e; B col l&31 same k;

@@ -3,7 +3,7 @@
 
 ``flash_attention(q, k, v, causal=...)`` runs the hand-written kernels in
 ``csrc/fa_fwd.hip`` / ``csrc/fa_bwd.hip`` — hardware-validated round 2
-(tests/test_fa_wip.py: fragment-layout probe + fwd/bwd numerics vs fp32
+(tests/test_fa.py: fragment-layout probe + fwd/bwd numerics vs fp32
 SDPA, all passing on MI355X).  ``attention()`` is the model-facing entry:
 it routes to the in-house kernels when the shapes qualify and the
 ``STOKE_USE_FA`` gate allows, else falls back to
